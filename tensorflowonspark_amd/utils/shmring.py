@@ -1,0 +1,137 @@
+"""Shared-memory block ring: the InputMode.SPARK bulk-data path.
+
+The reference pushed every RDD row through a ``multiprocessing.Manager`` proxy
+queue — two picklings and two socket hops *per sample* (reference
+``TFSparkNode.py:500-502`` producer, ``TFNode.py:279`` consumer); that is its
+known feed bottleneck. Here the payload travels through POSIX shared memory in
+*blocks* of many rows, and only tiny slot descriptors ride the manager queues
+(which also preserves the reference's ``queue.join()`` completion semantics —
+the feeder's ``join()`` returns only when the consumer has ``task_done()``-ed
+every block descriptor).
+
+Layout: one SharedMemory segment of ``slots * slot_bytes``. Producer:
+``slot = ring.acquire()`` (from the ``free`` queue) → write payload →
+``ring.commit(slot, n)`` (descriptor onto the ``data`` queue). Consumer:
+``ring.take()`` → read payload → ``ring.release(slot)``.
+
+On a GPU worker the consumer copies each block straight into a *pinned* staging
+tensor and launches ``hipMemcpyAsync`` on a side stream (see
+``TFNode.DataFeed.next_batch``), overlapping H2D with compute.
+"""
+
+import pickle
+from multiprocessing import shared_memory
+
+
+def _untrack(shm):
+    """Detach a SharedMemory from the resource tracker.
+
+    Attaching processes are not owners; without this, every attacher's
+    resource_tracker unlinks the segment at interpreter exit (CPython < 3.13
+    has no ``track=False``).
+    """
+    try:
+        from multiprocessing import resource_tracker
+        resource_tracker.unregister(shm._name, "shared_memory")
+    except Exception:
+        pass
+
+
+class BlockRing:
+    """A ring of fixed-size shared-memory slots coordinated by two queues."""
+
+    def __init__(self, name, slots, slot_bytes, data_queue, free_queue, create=False):
+        self.slots = slots
+        self.slot_bytes = slot_bytes
+        self._data_q = data_queue
+        self._free_q = free_queue
+        if create:
+            self.shm = shared_memory.SharedMemory(
+                name=name, create=True, size=slots * slot_bytes)
+            for i in range(slots):
+                free_queue.put(i)
+        else:
+            self.shm = shared_memory.SharedMemory(name=name)
+            _untrack(self.shm)
+        self.name = self.shm.name
+        self._owner = create
+
+    # -- producer side -------------------------------------------------------
+
+    def acquire(self, timeout=None):
+        """Block until a free slot index is available."""
+        return self._free_q.get(timeout=timeout) if timeout else self._free_q.get()
+
+    def write(self, slot, payload):
+        """Write raw bytes into a slot; returns length."""
+        n = len(payload)
+        if n > self.slot_bytes:
+            raise ValueError(
+                "payload {} B exceeds slot size {} B".format(n, self.slot_bytes))
+        off = slot * self.slot_bytes
+        self.shm.buf[off:off + n] = payload
+        return n
+
+    def commit(self, slot, nbytes, meta=None):
+        """Publish a written slot to the consumer."""
+        self._data_q.put((slot, nbytes, meta))
+        # 'free' queue join-semantics: mark our own get() as processed so the
+        # ring can be joined on the free queue as well if ever needed.
+        self._free_q.task_done()
+
+    def put_block(self, payload, meta=None, timeout=None):
+        """acquire + write + commit in one call."""
+        slot = self.acquire(timeout=timeout)
+        n = self.write(slot, payload)
+        self.commit(slot, n, meta)
+
+    def put_rows(self, rows, meta=None):
+        """Pickle a list of rows as one block (single pickling per block)."""
+        self.put_block(pickle.dumps(rows, protocol=pickle.HIGHEST_PROTOCOL), meta=meta)
+
+    # -- consumer side -------------------------------------------------------
+
+    def take(self, timeout=None):
+        """Get (slot, nbytes, meta) descriptor; None descriptor = end-of-feed."""
+        desc = self._data_q.get(timeout=timeout) if timeout else self._data_q.get()
+        return desc
+
+    def read(self, slot, nbytes):
+        """Return a *copy* of the slot payload as bytes."""
+        off = slot * self.slot_bytes
+        return bytes(self.shm.buf[off:off + nbytes])
+
+    def read_into(self, slot, nbytes, dest):
+        """Copy slot payload into a writable buffer (e.g. a pinned tensor)."""
+        off = slot * self.slot_bytes
+        dest[:nbytes] = self.shm.buf[off:off + nbytes]
+
+    def release(self, slot):
+        """Return the slot to the free pool and ack the descriptor."""
+        self._free_q.put(slot)
+        self._data_q.task_done()
+
+    def take_rows(self, timeout=None):
+        """Consume one pickled-rows block; returns (rows, meta) or (None, meta)."""
+        desc = self.take(timeout=timeout)
+        if desc is None:
+            return None, None
+        slot, nbytes, meta = desc
+        rows = pickle.loads(self.read(slot, nbytes))
+        self.release(slot)
+        return rows, meta
+
+    # -- lifecycle -----------------------------------------------------------
+
+    def close(self):
+        try:
+            self.shm.close()
+        except Exception:
+            pass
+
+    def unlink(self):
+        if self._owner:
+            try:
+                self.shm.unlink()
+            except Exception:
+                pass

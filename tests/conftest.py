@@ -1,0 +1,18 @@
+import multiprocessing
+import sys
+import os
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an AMD GPU (run on an MI355X box)")
+
+
+# fork is required: tests exercise multi-process managers/rings whose state is
+# inherited at fork time.
+try:
+    multiprocessing.set_start_method("fork", force=False)
+except RuntimeError:
+    pass
