@@ -158,8 +158,14 @@ class TFNodeContext:
 
 
 class TFSparkNode(object):
-    """Per-python-worker singleton state (reference ``TFSparkNode.py:111-125``)."""
+    """Per-python-worker singleton state (reference ``TFSparkNode.py:111-125``).
+
+    ``owned_mgr`` pins the handle returned by ``TFManager.start`` for the life
+    of the executor process: dropping the last reference to an *owning*
+    BaseManager runs its finalizer, which terminates the manager server — a
+    hazard when bootstrap and feeder tasks share one executor process."""
     mgr = None
+    owned_mgr = None
     cluster_id = None
 
 
@@ -291,6 +297,7 @@ def run(fn, tf_args, cluster_meta, tensorboard=False, log_dir=None, queues=None,
             mgr = TFManager.start(authkey, queue_names, "local")
         mgr.set("state", "running")
         TFSparkNode.mgr = mgr
+        TFSparkNode.owned_mgr = mgr
         TFSparkNode.cluster_id = cluster_id
 
         ring = None
@@ -614,6 +621,8 @@ def shutdown(cluster_info, queues=("input",), grace_secs=0):
 
         logger.info("shutting down executor %d", executor_id)
         for q in queues:
+            if q in ("error", "free"):
+                continue  # end-of-feed goes to data queues only
             try:
                 mgr.get_queue(q).put(None, block=True)
             except Exception as e:

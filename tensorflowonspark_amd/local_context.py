@@ -27,6 +27,8 @@ without a Spark install.
 
 import logging
 import multiprocessing
+
+import cloudpickle
 import os
 import queue as pyqueue
 import shutil
@@ -41,14 +43,23 @@ _STOP = "__stop__"
 
 
 def _executor_main(exec_id, workdir, task_q, result_q, env):
+    import cloudpickle
     os.makedirs(workdir, exist_ok=True)
     os.chdir(workdir)
     os.environ.update(env)
     while True:
         task = task_q.get()
         if task == _STOP:
+            # shut down any manager server this executor owns so no orphan
+            # process outlives the context (it would hold stdio pipes open)
+            try:
+                from .TFSparkNode import TFSparkNode
+                if TFSparkNode.owned_mgr is not None:
+                    TFSparkNode.owned_mgr.shutdown()
+            except Exception:
+                pass
             break
-        job_id, pid, fn, data, collect = task
+        job_id, pid, fn, data, collect = cloudpickle.loads(task)
         try:
             out = fn(iter(data))
             payload = list(out) if collect and out is not None else None
@@ -213,8 +224,8 @@ class LocalSparkContext:
                     "pinned job needs {} executors, have {}".format(
                         len(partitions), self.num_executors)
                 for pid in list(job.pending):
-                    self._task_qs[pid].put(
-                        (job.job_id, pid, fn, partitions[pid], collect))
+                    self._task_qs[pid].put(cloudpickle.dumps(
+                        (job.job_id, pid, fn, partitions[pid], collect)))
                     self._busy[pid] += 1
                 job.pending = []
             else:
@@ -267,8 +278,8 @@ class LocalSparkContext:
                 continue
             eid = idle.pop()
             pid = job.pending.pop(0)
-            self._task_qs[eid].put(
-                (job.job_id, pid, job.fn, job.partitions[pid], job.collect))
+            self._task_qs[eid].put(cloudpickle.dumps(
+                (job.job_id, pid, job.fn, job.partitions[pid], job.collect)))
             self._busy[eid] += 1
             if not job.pending:
                 self._job_order.pop(0)

@@ -1,0 +1,99 @@
+"""End-to-end cluster lifecycle on the Spark-free local executor pool
+(shape parity: reference tests/test_TFCluster.py — independent instances,
+InputMode.SPARK round trip, error propagation during and after feeding)."""
+
+import pytest
+
+from tensorflowonspark_amd import TFCluster
+from tensorflowonspark_amd.local_context import LocalSparkContext
+
+
+@pytest.fixture()
+def sc():
+    ctx = LocalSparkContext(num_executors=2)
+    yield ctx
+    ctx.stop()
+
+
+def _square_fn(args, ctx):
+    """Consume ints from the feed, square on the 'GPU worker', return results."""
+    feed = ctx.get_data_feed(train_mode=False)
+    while not feed.should_stop():
+        batch = feed.next_batch(10)
+        if not batch:
+            break
+        feed.batch_results([x * x for x in batch])
+
+
+def test_inference_roundtrip(sc):
+    cluster = TFCluster.run(sc, _square_fn, {}, num_executors=2, num_ps=0,
+                            master_node=None, input_mode=TFCluster.InputMode.SPARK,
+                            num_gpus=0, reservation_timeout=60)
+    data = list(range(1000))
+    rdd = sc.parallelize(data, 4)
+    out = cluster.inference(rdd).collect()
+    assert sorted(out) == sorted(x * x for x in data)
+    cluster.shutdown(grace_secs=0)
+
+
+def _sum_train_fn(args, ctx):
+    feed = ctx.get_data_feed(train_mode=True)
+    total = 0
+    while not feed.should_stop():
+        batch = feed.next_batch(100)
+        total += sum(batch)
+    # write result where the test can see it (executor cwd)
+    with open("sum_result.txt", "w") as f:
+        f.write(str(total))
+
+
+def test_train_feed(sc):
+    cluster = TFCluster.run(sc, _sum_train_fn, {}, num_executors=2, num_ps=0,
+                            master_node=None, input_mode=TFCluster.InputMode.SPARK,
+                            num_gpus=0, reservation_timeout=60)
+    data = list(range(100))
+    rdd = sc.parallelize(data, 2)
+    cluster.train(rdd, num_epochs=2)
+    cluster.shutdown(grace_secs=1)
+    # both executors together consumed 2 epochs of the data
+    import glob
+    import os
+    total = 0
+    for f in glob.glob(os.path.join(sc._root, "executor_*", "sum_result.txt")):
+        with open(f) as fh:
+            total += int(fh.read())
+    assert total == 2 * sum(data)
+
+
+def _failing_fn(args, ctx):
+    feed = ctx.get_data_feed(train_mode=True)
+    feed.next_batch(1)
+    raise RuntimeError("injected worker failure")
+
+
+def test_error_during_feeding(sc):
+    cluster = TFCluster.run(sc, _failing_fn, {}, num_executors=2, num_ps=0,
+                            master_node=None, input_mode=TFCluster.InputMode.SPARK,
+                            num_gpus=0, reservation_timeout=60)
+    rdd = sc.parallelize(range(1000), 2)
+    with pytest.raises(Exception, match="injected worker failure"):
+        cluster.train(rdd, num_epochs=5, feed_timeout=30)
+        cluster.shutdown(grace_secs=0)
+
+
+def _late_failing_fn(args, ctx):
+    feed = ctx.get_data_feed(train_mode=True)
+    while not feed.should_stop():
+        if not feed.next_batch(100):
+            break
+    raise RuntimeError("late failure after feeding")
+
+
+def test_error_after_feeding_caught_by_shutdown(sc):
+    cluster = TFCluster.run(sc, _late_failing_fn, {}, num_executors=2, num_ps=0,
+                            master_node=None, input_mode=TFCluster.InputMode.SPARK,
+                            num_gpus=0, reservation_timeout=60)
+    rdd = sc.parallelize(range(100), 2)
+    cluster.train(rdd, num_epochs=1)
+    with pytest.raises(Exception, match="late failure"):
+        cluster.shutdown(grace_secs=3)
