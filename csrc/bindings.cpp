@@ -1,0 +1,233 @@
+// Python bindings for the tfosr gfx950 kernel set. Compiled with hipcc;
+// kernels live in the .hip translation units and are reached through the
+// extern "C" launchers so torch headers stay out of kernel code.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void tfosr_bn_stats(const void*, int, int, float*, float*, int, int, long, hipStream_t);
+void tfosr_bn_finalize(const float*, const float*, float*, float*, float*, float*,
+                       long, int, float, float, hipStream_t);
+void tfosr_bn_relu_apply(const void*, void*, const float*, const float*,
+                         const float*, const float*, int, int, long, int, long,
+                         hipStream_t);
+void tfosr_bn_bwd_stats(const void*, const void*, const void*, const float*,
+                        const float*, float*, float*, int, int, int, int, long,
+                        hipStream_t);
+void tfosr_bn_bwd_dx(const void*, const void*, const void*, const float*,
+                     const float*, const float*, const float*, const float*,
+                     void*, int, int, long, int, long, hipStream_t);
+void tfosr_xent_fwd(const void*, const long*, float*, float*, int, int, int,
+                    hipStream_t);
+void tfosr_xent_bwd(const void*, const float*, const long*, const float*, void*,
+                    int, int, int, hipStream_t);
+void tfosr_nhwc_pack(const void*, void*, const float*, const float*, float, int,
+                     int, long, int, long, hipStream_t);
+void tfosr_sgd_step(float*, const float*, float*, float, float, float, int, long,
+                    hipStream_t);
+void tfosr_gemm_bt(const void*, const void*, void*, int, int, int, int,
+                   hipStream_t);
+void tfosr_mfma_probe(const short*, const short*, float*, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+struct BNLayout {
+  bool nhwc;
+  int N, C;
+  long HW;
+};
+
+BNLayout bn_layout(const at::Tensor& x) {
+  TORCH_CHECK(x.dim() == 4, "expected 4-D NCHW tensor");
+  BNLayout l;
+  l.N = x.size(0);
+  l.C = x.size(1);
+  l.HW = (long)x.size(2) * x.size(3);
+  if (x.is_contiguous(at::MemoryFormat::ChannelsLast)) {
+    l.nhwc = true;
+  } else {
+    TORCH_CHECK(x.is_contiguous(), "input must be contiguous (NCHW or channels_last)");
+    l.nhwc = false;
+  }
+  return l;
+}
+
+int dtype_flag(const at::Tensor& x) {
+  if (x.scalar_type() == at::kBFloat16) return 1;
+  TORCH_CHECK(x.scalar_type() == at::kFloat, "expected float32 or bfloat16");
+  return 0;
+}
+
+std::vector<at::Tensor> bn_relu_fwd_train(at::Tensor x, at::Tensor w, at::Tensor b,
+                                          at::Tensor rm, at::Tensor rv,
+                                          double momentum, double eps) {
+  auto l = bn_layout(x);
+  int bf = dtype_flag(x);
+  auto opts = x.options().dtype(at::kFloat);
+  auto wsum = at::zeros({l.C}, opts);
+  auto wsq = at::zeros({l.C}, opts);
+  auto save_mean = at::empty({l.C}, opts);
+  auto save_rstd = at::empty({l.C}, opts);
+  auto y = at::empty_like(x);
+  auto s = cur_stream();
+  long M = (long)l.N * l.HW;
+  tfosr_bn_stats(x.data_ptr(), bf, l.nhwc, wsum.data_ptr<float>(),
+                 wsq.data_ptr<float>(), l.N, l.C, l.HW, s);
+  tfosr_bn_finalize(wsum.data_ptr<float>(), wsq.data_ptr<float>(),
+                    save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
+                    rm.data_ptr<float>(), rv.data_ptr<float>(), M, l.C,
+                    (float)momentum, (float)eps, s);
+  tfosr_bn_relu_apply(x.data_ptr(), y.data_ptr(), save_mean.data_ptr<float>(),
+                      save_rstd.data_ptr<float>(), w.data_ptr<float>(),
+                      b.data_ptr<float>(), bf, l.nhwc, M * l.C, l.C, l.HW, s);
+  return {y, save_mean, save_rstd};
+}
+
+at::Tensor bn_relu_fwd_eval(at::Tensor x, at::Tensor w, at::Tensor b,
+                            at::Tensor rm, at::Tensor rv, double eps) {
+  auto l = bn_layout(x);
+  int bf = dtype_flag(x);
+  auto rstd = at::rsqrt(rv + eps);
+  auto y = at::empty_like(x);
+  tfosr_bn_relu_apply(x.data_ptr(), y.data_ptr(), rm.data_ptr<float>(),
+                      rstd.data_ptr<float>(), w.data_ptr<float>(),
+                      b.data_ptr<float>(), bf, l.nhwc,
+                      (long)l.N * l.HW * l.C, l.C, l.HW, cur_stream());
+  return y;
+}
+
+std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
+                                    at::Tensor w, at::Tensor save_mean,
+                                    at::Tensor save_rstd) {
+  auto l = bn_layout(x);
+  int bf = dtype_flag(x);
+  if (l.nhwc) {
+    dy = dy.contiguous(at::MemoryFormat::ChannelsLast);
+  } else {
+    dy = dy.contiguous();
+  }
+  auto opts = x.options().dtype(at::kFloat);
+  auto dg = at::zeros({l.C}, opts);
+  auto db = at::zeros({l.C}, opts);
+  auto dx = at::empty_like(x);
+  auto s = cur_stream();
+  tfosr_bn_bwd_stats(x.data_ptr(), dy.data_ptr(), y.data_ptr(),
+                     save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
+                     dg.data_ptr<float>(), db.data_ptr<float>(), bf, l.nhwc,
+                     l.N, l.C, l.HW, s);
+  tfosr_bn_bwd_dx(x.data_ptr(), dy.data_ptr(), y.data_ptr(),
+                  save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
+                  w.data_ptr<float>(), dg.data_ptr<float>(), db.data_ptr<float>(),
+                  dx.data_ptr(), bf, l.nhwc, (long)l.N * l.HW * l.C, l.C, l.HW, s);
+  // dgamma/dbeta are the reduced sums (save_* already folded into dg)
+  return {dx, dg, db};
+}
+
+std::vector<at::Tensor> softmax_xent_fwd(at::Tensor logits, at::Tensor target) {
+  TORCH_CHECK(logits.dim() == 2 && logits.is_contiguous());
+  TORCH_CHECK(target.scalar_type() == at::kLong);
+  int N = logits.size(0), C = logits.size(1);
+  auto opts = logits.options().dtype(at::kFloat);
+  auto loss = at::empty({N}, opts);
+  auto lse = at::empty({N}, opts);
+  tfosr_xent_fwd(logits.data_ptr(), target.data_ptr<long>(),
+                 loss.data_ptr<float>(), lse.data_ptr<float>(),
+                 dtype_flag(logits), N, C, cur_stream());
+  return {loss, lse};
+}
+
+at::Tensor softmax_xent_bwd(at::Tensor logits, at::Tensor lse, at::Tensor target,
+                            at::Tensor gout) {
+  int N = logits.size(0), C = logits.size(1);
+  auto dlogits = at::empty_like(logits);
+  tfosr_xent_bwd(logits.data_ptr(), lse.data_ptr<float>(),
+                 target.data_ptr<long>(), gout.contiguous().data_ptr<float>(),
+                 dlogits.data_ptr(), dtype_flag(logits), N, C, cur_stream());
+  return dlogits;
+}
+
+at::Tensor nhwc_pack(at::Tensor x, at::Tensor mean, at::Tensor std_, double scale,
+                     bool out_bf16, bool channels_last) {
+  TORCH_CHECK(x.dim() == 4 && x.scalar_type() == at::kByte && x.is_contiguous(),
+              "nhwc_pack expects contiguous uint8 NHWC");
+  int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  long HW = (long)H * W;
+  auto opts = x.options().dtype(out_bf16 ? at::kBFloat16 : at::kFloat);
+  at::Tensor out;
+  if (channels_last) {
+    out = at::empty({N, C, H, W}, opts, at::MemoryFormat::ChannelsLast);
+  } else {
+    out = at::empty({N, C, H, W}, opts);
+  }
+  tfosr_nhwc_pack(x.data_ptr(), out.data_ptr(), mean.data_ptr<float>(),
+                  std_.data_ptr<float>(), (float)scale, out_bf16, channels_last,
+                  (long)N * C * HW, C, HW, cur_stream());
+  return out;
+}
+
+void sgd_step(at::Tensor p, at::Tensor g, at::Tensor m, double lr, double mu,
+              double wd, bool nesterov) {
+  TORCH_CHECK(p.is_contiguous() && g.is_contiguous() && m.is_contiguous());
+  TORCH_CHECK(p.scalar_type() == at::kFloat, "sgd_step expects fp32 params");
+  tfosr_sgd_step(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+                 (float)lr, (float)mu, (float)wd, nesterov, p.numel(),
+                 cur_stream());
+}
+
+at::Tensor gemm_bt(at::Tensor a, at::Tensor b, bool out_bf16) {
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1),
+              "gemm_bt: A[M,K], B[N,K]");
+  TORCH_CHECK(a.scalar_type() == at::kBFloat16 && b.scalar_type() == at::kBFloat16);
+  int M = a.size(0), N = b.size(0), K = a.size(1);
+  if (K % 32 != 0) {
+    int Kp = (K + 31) / 32 * 32;
+    auto ap = at::zeros({M, Kp}, a.options());
+    auto bp = at::zeros({N, Kp}, b.options());
+    ap.narrow(1, 0, K).copy_(a);
+    bp.narrow(1, 0, K).copy_(b);
+    a = ap; b = bp; K = Kp;
+  }
+  a = a.contiguous();
+  b = b.contiguous();
+  auto out = at::empty({M, N}, a.options().dtype(out_bf16 ? at::kBFloat16
+                                                          : at::kFloat));
+  tfosr_gemm_bt(a.data_ptr(), b.data_ptr(), out.data_ptr(), out_bf16, M, N, K,
+                cur_stream());
+  return out;
+}
+
+at::Tensor gemm_bf16(at::Tensor a, at::Tensor b) {
+  // A[M,K] @ B[K,N]: feed B^T (contiguous [N,K]) to the kernel
+  return gemm_bt(a, b.t().contiguous(), /*out_bf16=*/false);
+}
+
+at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
+  TORCH_CHECK(a.sizes() == at::IntArrayRef({64, 8}) &&
+              b.sizes() == at::IntArrayRef({64, 8}));
+  auto c = at::empty({64, 4}, a.options().dtype(at::kFloat));
+  tfosr_mfma_probe(a.contiguous().data_ptr<short>(),
+                   b.contiguous().data_ptr<short>(), c.data_ptr<float>(),
+                   cur_stream());
+  return c;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bn_relu_fwd_train", &bn_relu_fwd_train);
+  m.def("bn_relu_fwd_eval", &bn_relu_fwd_eval);
+  m.def("bn_relu_bwd", &bn_relu_bwd);
+  m.def("softmax_xent_fwd", &softmax_xent_fwd);
+  m.def("softmax_xent_bwd", &softmax_xent_bwd);
+  m.def("nhwc_pack", &nhwc_pack);
+  m.def("sgd_step", &sgd_step);
+  m.def("gemm_bt", &gemm_bt, py::arg("a"), py::arg("b"), py::arg("out_bf16") = false);
+  m.def("gemm_bf16", &gemm_bf16);
+  m.def("mfma_probe", &mfma_probe);
+}

@@ -1,0 +1,123 @@
+// Elementwise / optimizer kernels for gfx950:
+//  - nhwc_pack: uint8 NHWC batch -> normalized NCHW (or channels_last) f32/bf16
+//  - sgd_step: fused SGD(momentum, weight_decay) over a flat parameter bucket
+#include "tfosr_common.h"
+
+// ---------------------------------------------------------------------------
+// NHWC uint8 -> normalized tensor (the DataFeed->GPU ingest kernel)
+// ---------------------------------------------------------------------------
+
+typedef uint8_t u8;
+typedef u8 u8x16 __attribute__((ext_vector_type(16)));
+
+// channels_last output: pure elementwise in source order (fast path).
+template <typename T>
+__global__ void nhwc_pack_cl_kernel(const u8* __restrict__ in, T* __restrict__ out,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ std_,
+                                    float scale, long total, int C) {
+  constexpr int V = 16;
+  const long nvec = total / V;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    const long base = i * V;
+    u8x16 v = *(const u8x16*)(in + base);
+    T o[V];
+    #pragma unroll
+    for (int j = 0; j < V; ++j) {
+      int c = (int)((base + j) % C);
+      o[j] = (T)(((float)v[j] * scale - mean[c]) / std_[c]);
+    }
+    #pragma unroll
+    for (int j = 0; j < V; ++j) out[base + j] = o[j];
+  }
+  long tail0 = nvec * V;
+  for (long idx = tail0 + blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    int c = (int)(idx % C);
+    out[idx] = (T)(((float)in[idx] * scale - mean[c]) / std_[c]);
+  }
+}
+
+// NCHW-contiguous output: iterate output index, gather strided u8 reads
+// (u8 gather is absorbed by L2; output writes stay coalesced).
+template <typename T>
+__global__ void nhwc_pack_nchw_kernel(const u8* __restrict__ in, T* __restrict__ out,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ std_,
+                                      float scale, long total, int C, long HW) {
+  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    long n = idx / (C * HW);
+    int c = (int)((idx / HW) % C);
+    long hw = idx % HW;
+    u8 v = in[(n * HW + hw) * C + c];
+    out[idx] = (T)(((float)v * scale - mean[c]) / std_[c]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused flat SGD with momentum:  m = mu*m + g + wd*p;  p -= lr * (m or g+mu*m)
+// ---------------------------------------------------------------------------
+
+__global__ void sgd_step_kernel(float* __restrict__ p, const float* __restrict__ g,
+                                float* __restrict__ m, float lr, float mu,
+                                float wd, int nesterov, long n) {
+  constexpr int V = 4;
+  const long nvec = n / V;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    f4v pv = *(f4v*)(p + i * V);
+    f4v gv = *(const f4v*)(g + i * V);
+    f4v mv = *(f4v*)(m + i * V);
+    #pragma unroll
+    for (int j = 0; j < V; ++j) {
+      float grad = gv[j] + wd * pv[j];
+      mv[j] = mu * mv[j] + grad;
+      float upd = nesterov ? grad + mu * mv[j] : mv[j];
+      pv[j] -= lr * upd;
+    }
+    *(f4v*)(m + i * V) = mv;
+    *(f4v*)(p + i * V) = pv;
+  }
+  for (long i = nvec * V + blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float grad = g[i] + wd * p[i];
+    m[i] = mu * m[i] + grad;
+    p[i] -= lr * (nesterov ? grad + mu * m[i] : m[i]);
+  }
+}
+
+extern "C" {
+
+void tfosr_nhwc_pack(const void* in, void* out, const float* mean,
+                     const float* std_, float scale, int out_bf16,
+                     int channels_last, long total, int C, long HW,
+                     hipStream_t s) {
+  int grid = tfosr_grid(total / 16, 256);
+  if (channels_last) {
+    if (out_bf16)
+      hipLaunchKernelGGL(nhwc_pack_cl_kernel<bf16_t>, dim3(grid), dim3(256), 0, s,
+                         (const u8*)in, (bf16_t*)out, mean, std_, scale, total, C);
+    else
+      hipLaunchKernelGGL(nhwc_pack_cl_kernel<float>, dim3(grid), dim3(256), 0, s,
+                         (const u8*)in, (float*)out, mean, std_, scale, total, C);
+  } else {
+    grid = tfosr_grid(total, 256);
+    if (out_bf16)
+      hipLaunchKernelGGL(nhwc_pack_nchw_kernel<bf16_t>, dim3(grid), dim3(256), 0, s,
+                         (const u8*)in, (bf16_t*)out, mean, std_, scale, total, C, HW);
+    else
+      hipLaunchKernelGGL(nhwc_pack_nchw_kernel<float>, dim3(grid), dim3(256), 0, s,
+                         (const u8*)in, (float*)out, mean, std_, scale, total, C, HW);
+  }
+}
+
+void tfosr_sgd_step(float* p, const float* g, float* m, float lr, float mu,
+                    float wd, int nesterov, long n, hipStream_t s) {
+  int grid = tfosr_grid(n / 4, 256);
+  hipLaunchKernelGGL(sgd_step_kernel, dim3(grid), dim3(256), 0, s,
+                     p, g, m, lr, mu, wd, nesterov, n);
+}
+
+}  // extern "C"

@@ -159,6 +159,33 @@ class DataFeed(object):
                 cols[name].append(row[i])
         return cols
 
+    def next_arrays(self):
+        """Consume one array block fed via ``BlockRing.put_arrays``.
+
+        Returns {name: np.ndarray} (views copied out of the ring), or None at
+        end-of-feed. Partition markers are transparent here.
+        """
+        while True:
+            item = self._queue_in.get(block=True)
+            if item is None:
+                self.done_feeding = True
+                self._queue_in.task_done()
+                return None
+            kind = item[0]
+            if kind == "end_partition":
+                self._queue_in.task_done()
+                continue
+            if kind == "shm_arr":
+                _, slot, _nbytes, meta = item
+                ring = self._attach_ring()
+                out = ring.read_arrays(slot, meta, copy=True)
+                ring._free_q.put(slot)
+                self._queue_in.task_done()
+                return out
+            raise RuntimeError(
+                "next_arrays got a non-array item {!r}; mixed feeds should use "
+                "next_batch".format(item[0]))
+
     def should_stop(self):
         return self.done_feeding
 

@@ -1,0 +1,2 @@
+from .mnist import MNISTNet, MNISTMLP  # noqa: F401
+from .resnet import ResNet, resnet50, resnet56_cifar  # noqa: F401

@@ -1,0 +1,138 @@
+"""ResNet family (ResNet-50 v1.5 ImageNet; ResNet-56 CIFAR).
+
+The reference's benchmark workload is the TF-official ResNet run under TFoS
+(reference ``examples/resnet/resnet_cifar_dist.py``; the synthetic-data path at
+``:160-168`` is the baseline template). This is a from-scratch PyTorch
+implementation of the same architectures, structured so the BN+ReLU pairs route
+through the fused CDNA4 HIP kernel (``ops.FusedBNReLU``) on GPU and through
+plain PyTorch on CPU.
+
+ResNet-50 v1.5: the stride-2 3x3 conv sits in the middle of the bottleneck
+(not the 1x1), matching the config every published ResNet-50 images/sec number
+uses.
+"""
+
+import torch
+import torch.nn as nn
+
+from ..ops.modules import FusedBNReLU
+
+
+def conv3x3(cin, cout, stride=1):
+    return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
+
+
+def conv1x1(cin, cout, stride=1):
+    return nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
+
+
+class Bottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, cin, width, stride=1, downsample=None):
+        super().__init__()
+        self.conv1 = conv1x1(cin, width)
+        self.bnrelu1 = FusedBNReLU(width)
+        self.conv2 = conv3x3(width, width, stride)  # v1.5: stride on the 3x3
+        self.bnrelu2 = FusedBNReLU(width)
+        self.conv3 = conv1x1(width, width * self.expansion)
+        self.bn3 = nn.BatchNorm2d(width * self.expansion)
+        self.relu = nn.ReLU(inplace=True)
+        self.downsample = downsample
+
+    def forward(self, x):
+        identity = x
+        out = self.bnrelu1(self.conv1(x))
+        out = self.bnrelu2(self.conv2(out))
+        out = self.bn3(self.conv3(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        out += identity
+        return self.relu(out)
+
+
+class BasicBlock(nn.Module):
+    expansion = 1
+
+    def __init__(self, cin, width, stride=1, downsample=None):
+        super().__init__()
+        self.conv1 = conv3x3(cin, width, stride)
+        self.bnrelu1 = FusedBNReLU(width)
+        self.conv2 = conv3x3(width, width)
+        self.bn2 = nn.BatchNorm2d(width)
+        self.relu = nn.ReLU(inplace=True)
+        self.downsample = downsample
+
+    def forward(self, x):
+        identity = x
+        out = self.bnrelu1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        out += identity
+        return self.relu(out)
+
+
+class ResNet(nn.Module):
+    def __init__(self, block, layers, num_classes=1000, cifar_stem=False,
+                 base_width=64):
+        super().__init__()
+        self.inplanes = 16 if cifar_stem else base_width
+        if cifar_stem:
+            self.stem = nn.Sequential(
+                conv3x3(3, self.inplanes), FusedBNReLU(self.inplanes))
+            widths = [16, 32, 64]
+            strides = [1, 2, 2]
+        else:
+            self.stem = nn.Sequential(
+                nn.Conv2d(3, self.inplanes, 7, stride=2, padding=3, bias=False),
+                FusedBNReLU(self.inplanes),
+                nn.MaxPool2d(3, stride=2, padding=1))
+            widths = [64, 128, 256, 512]
+            strides = [1, 2, 2, 2]
+        stages = []
+        for w, s, n in zip(widths, strides, layers):
+            stages.append(self._make_stage(block, w, n, s))
+        self.stages = nn.Sequential(*stages)
+        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self.fc = nn.Linear(widths[-1] * block.expansion, num_classes)
+
+        for m in self.modules():
+            if isinstance(m, nn.Conv2d):
+                nn.init.kaiming_normal_(m.weight, mode="fan_out", nonlinearity="relu")
+            elif isinstance(m, nn.BatchNorm2d):
+                nn.init.ones_(m.weight)
+                nn.init.zeros_(m.bias)
+        # zero-init last BN in each block (standard ResNet recipe)
+        for m in self.modules():
+            if isinstance(m, Bottleneck):
+                nn.init.zeros_(m.bn3.weight)
+            elif isinstance(m, BasicBlock):
+                nn.init.zeros_(m.bn2.weight)
+
+    def _make_stage(self, block, width, blocks, stride):
+        downsample = None
+        if stride != 1 or self.inplanes != width * block.expansion:
+            downsample = nn.Sequential(
+                conv1x1(self.inplanes, width * block.expansion, stride),
+                nn.BatchNorm2d(width * block.expansion))
+        layers = [block(self.inplanes, width, stride, downsample)]
+        self.inplanes = width * block.expansion
+        for _ in range(1, blocks):
+            layers.append(block(self.inplanes, width))
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        x = self.stem(x)
+        x = self.stages(x)
+        x = self.avgpool(x)
+        x = torch.flatten(x, 1)
+        return self.fc(x)
+
+
+def resnet50(num_classes=1000):
+    return ResNet(Bottleneck, [3, 4, 6, 3], num_classes=num_classes)
+
+
+def resnet56_cifar(num_classes=10):
+    return ResNet(BasicBlock, [9, 9, 9], num_classes=num_classes, cifar_stem=True)

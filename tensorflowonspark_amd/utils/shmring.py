@@ -89,6 +89,43 @@ class BlockRing:
         """Pickle a list of rows as one block (single pickling per block)."""
         self.put_block(pickle.dumps(rows, protocol=pickle.HIGHEST_PROTOCOL), meta=meta)
 
+    def put_arrays(self, arrays, timeout=None):
+        """Zero-pickle block of named numpy arrays (the fast tensor-feed path).
+
+        Payload = raw array bytes back to back; the descriptor meta carries
+        (name, dtype, shape, nbytes) per array. The consumer can read each
+        array straight into a pinned staging tensor (``read_into``) for
+        ``hipMemcpyAsync`` H2D.
+        """
+        slot = self.acquire(timeout=timeout)
+        off = slot * self.slot_bytes
+        meta = []
+        pos = 0
+        for name, arr in arrays.items():
+            b = arr.tobytes() if not arr.flags["C_CONTIGUOUS"] else memoryview(arr).cast("B")
+            n = len(b)
+            if pos + n > self.slot_bytes:
+                raise ValueError("array block exceeds slot size")
+            self.shm.buf[off + pos:off + pos + n] = b
+            meta.append((name, arr.dtype.str, arr.shape, n))
+            pos += n
+        self._data_q.put(("shm_arr", slot, pos, meta))
+        self._free_q.task_done()
+
+    def read_arrays(self, slot, meta, copy=True):
+        """Materialize a put_arrays block as {name: np.ndarray}."""
+        import numpy as np
+        off = slot * self.slot_bytes
+        out = {}
+        pos = 0
+        for name, dtype, shape, n in meta:
+            view = np.frombuffer(self.shm.buf, dtype=np.dtype(dtype),
+                                 count=int(np.prod(shape)) if shape else 1,
+                                 offset=off + pos).reshape(shape)
+            out[name] = view.copy() if copy else view
+            pos += n
+        return out
+
     # -- consumer side -------------------------------------------------------
 
     def take(self, timeout=None):
