@@ -98,27 +98,60 @@ class SparkFeed:
         self.feed = TFNode.DataFeed(self.mgr, train_mode=True)
         self.device = device
         self.use_cuda = device.type == "cuda"
+        self._stop = False
         if self.use_cuda:
-            self.x_pin = torch.empty((batch,) + shape, dtype=torch.uint8,
-                                     pin_memory=True)
-            self.y_pin = torch.empty((batch,), dtype=torch.int64, pin_memory=True)
+            # double-buffered pinned staging + background prefetch thread:
+            # shm -> pinned (thread) overlaps compute; pinned -> HBM is an
+            # async hipMemcpyAsync on a side stream
+            import queue as pyq
+            import threading
+            self.nbuf = 3
+            self.x_pin = [torch.empty((batch,) + shape, dtype=torch.uint8,
+                                      pin_memory=True) for _ in range(self.nbuf)]
+            self.y_pin = [torch.empty((batch,), dtype=torch.int64,
+                                      pin_memory=True) for _ in range(self.nbuf)]
             self.copy_stream = torch.cuda.Stream()
+            self._ready = pyq.Queue(maxsize=self.nbuf)
+            self._free = pyq.Queue()
+            for i in range(self.nbuf):
+                self._free.put((i, None))
+            self._thread = threading.Thread(target=self._prefetch_loop, daemon=True)
+            self._thread.start()
+
+    def _prefetch_loop(self):
+        while not self._stop:
+            i, evt = self._free.get()
+            if i is None:
+                break
+            if evt is not None:
+                evt.synchronize()  # prior H2D from this buffer must be done
+            dest = {"x": self.x_pin[i].numpy(), "y": self.y_pin[i].numpy()}
+            if not self.feed.next_arrays_into(dest):
+                self._ready.put(None)
+                break
+            self._ready.put(i)
 
     def next(self):
-        arrs = self.feed.next_arrays()
-        x_np, y_np = arrs["x"], arrs["y"]
         if self.use_cuda:
-            self.x_pin.copy_(torch.from_numpy(x_np))
-            self.y_pin.copy_(torch.from_numpy(y_np))
+            i = self._ready.get()
+            if i is None:
+                raise StopIteration("feed ended")
             with torch.cuda.stream(self.copy_stream):
-                x = self.x_pin.to(self.device, non_blocking=True)
-                y = self.y_pin.to(self.device, non_blocking=True)
+                x = self.x_pin[i].to(self.device, non_blocking=True)
+                y = self.y_pin[i].to(self.device, non_blocking=True)
+                evt = torch.cuda.Event()
+                evt.record(self.copy_stream)
             torch.cuda.current_stream().wait_stream(self.copy_stream)
+            self._free.put((i, evt))
             return x, y
-        return torch.from_numpy(x_np), torch.from_numpy(y_np)
+        arrs = self.feed.next_arrays()
+        return torch.from_numpy(arrs["x"]), torch.from_numpy(arrs["y"])
 
     def close(self):
+        self._stop = True
         self.stop_evt.set()
+        if self.use_cuda:
+            self._free.put((None, None))
         self.proc.join(timeout=10)
         if self.proc.is_alive():
             self.proc.terminate()

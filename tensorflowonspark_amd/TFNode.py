@@ -186,6 +186,37 @@ class DataFeed(object):
                 "next_arrays got a non-array item {!r}; mixed feeds should use "
                 "next_batch".format(item[0]))
 
+    def next_arrays_into(self, dest):
+        """Like ``next_arrays`` but copies each named array straight into a
+        caller-provided writable buffer (e.g. numpy views of *pinned* torch
+        tensors) — one memcpy from shared memory to pinned staging, ready for
+        hipMemcpyAsync. Returns True, or False at end-of-feed."""
+        while True:
+            item = self._queue_in.get(block=True)
+            if item is None:
+                self.done_feeding = True
+                self._queue_in.task_done()
+                return False
+            kind = item[0]
+            if kind == "end_partition":
+                self._queue_in.task_done()
+                continue
+            if kind == "shm_arr":
+                _, slot, _nbytes, meta = item
+                ring = self._attach_ring()
+                off = slot * ring.slot_bytes
+                pos = 0
+                for name, dtype, shape, n in meta:
+                    buf = dest[name]
+                    mv = memoryview(buf).cast("B")
+                    mv[:n] = ring.shm.buf[off + pos:off + pos + n]
+                    pos += n
+                ring._free_q.put(slot)
+                self._queue_in.task_done()
+                return True
+            raise RuntimeError(
+                "next_arrays_into got non-array item {!r}".format(item[0]))
+
     def should_stop(self):
         return self.done_feeding
 

@@ -90,8 +90,18 @@ def _busy_gpu_ids():
                 gpu_id = entry.get("gpu")
                 if gpu_id is None:
                     continue
-                real = [p for p in procs
-                        if isinstance(p, dict) and p.get("process_info") != "N/A" and p]
+                real = []
+                for p in procs:
+                    if not isinstance(p, dict):
+                        continue
+                    info = p.get("process_info")
+                    if info == "N/A" or info is None:
+                        continue
+                    # amd-smi emits a placeholder entry with name "N/A" when
+                    # no process is attached (observed on MI355X, ROCm 7.2)
+                    if isinstance(info, dict) and info.get("name") in ("N/A", None):
+                        continue
+                    real.append(p)
                 if real:
                     busy.add(int(gpu_id))
         except (ValueError, KeyError, TypeError):

@@ -1,0 +1,201 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32 reference.
+
+All tests @pytest.mark.gpu — they require an MI355X and the in-tree
+tfosr_hip_ops.so (ops raise if it's missing; no silent eager fallback).
+"""
+
+import pytest
+import torch
+
+gpu = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="needs a GPU")
+
+
+def _ext():
+    from tensorflowonspark_amd.ops import get_ext
+    e = get_ext(required=True)
+    assert e is not None
+    return e
+
+
+def _close(a, b, rtol, atol, what=""):
+    a = a.float().cpu()
+    b = b.float().cpu()
+    err = (a - b).abs().max().item()
+    ok = torch.allclose(a, b, rtol=rtol, atol=atol)
+    assert ok, "{}: max abs err {}".format(what, err)
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("channels_last", [False, True])
+def test_bn_relu_forward_train(dtype, channels_last):
+    ext = _ext()
+    torch.manual_seed(0)
+    N, C, H, W = 8, 32, 14, 14
+    x = torch.randn(N, C, H, W, device="cuda").to(dtype)
+    if channels_last:
+        x = x.contiguous(memory_format=torch.channels_last)
+    w = torch.rand(C, device="cuda") + 0.5
+    b = torch.randn(C, device="cuda")
+    rm = torch.zeros(C, device="cuda")
+    rv = torch.ones(C, device="cuda")
+    rm_ref, rv_ref = rm.clone(), rv.clone()
+
+    y, mean, rstd = ext.bn_relu_fwd_train(x, w, b, rm, rv, 0.1, 1e-5)
+
+    xf = x.float()
+    y_ref = torch.nn.functional.relu(torch.nn.functional.batch_norm(
+        xf, rm_ref, rv_ref, w, b, True, 0.1, 1e-5))
+    tol = 1e-4 if dtype == torch.float32 else 3e-2
+    _close(y, y_ref, 1e-2, tol, "bn_relu fwd y")
+    _close(rm, rm_ref, 1e-3, 1e-4, "running_mean")
+    _close(rv, rv_ref, 1e-2, 1e-3, "running_var")
+    _close(mean, xf.mean(dim=(0, 2, 3)), 1e-3, 1e-4, "save_mean")
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("channels_last", [False, True])
+def test_bn_relu_backward(dtype, channels_last):
+    from tensorflowonspark_amd.ops.modules import FusedBNReLU
+    torch.manual_seed(1)
+    N, C, H, W = 4, 16, 8, 8
+    x0 = torch.randn(N, C, H, W)
+
+    # reference on CPU fp32
+    ref = FusedBNReLU(C)
+    x_ref = x0.clone().requires_grad_(True)
+    y_ref = ref(x_ref)
+    gy = torch.randn_like(y_ref)
+    y_ref.backward(gy)
+
+    mod = FusedBNReLU(C).cuda()
+    mod.load_state_dict({k: v.cuda() for k, v in ref.state_dict().items()})
+    x = x0.to("cuda").to(dtype)
+    if channels_last:
+        x = x.contiguous(memory_format=torch.channels_last)
+    x.requires_grad_(True)
+    y = mod(x)
+    y.backward(gy.to("cuda").to(dtype))
+
+    tol = (1e-3, 1e-4) if dtype == torch.float32 else (5e-2, 5e-2)
+    _close(y, y_ref, *tol, "bn y")
+    _close(x.grad, x_ref.grad, *tol, "bn dx")
+    _close(mod.weight.grad, ref.weight.grad, *tol, "bn dgamma")
+    _close(mod.bias.grad, ref.bias.grad, *tol, "bn dbeta")
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_softmax_xent(dtype):
+    from tensorflowonspark_amd.ops.modules import softmax_cross_entropy
+    torch.manual_seed(2)
+    N, C = 64, 1000
+    logits0 = torch.randn(N, C) * 3
+    target = torch.randint(0, C, (N,))
+
+    ref_l = logits0.clone().requires_grad_(True)
+    loss_ref = torch.nn.functional.cross_entropy(ref_l, target)
+    loss_ref.backward()
+
+    lg = logits0.to("cuda").to(dtype).requires_grad_(True)
+    loss = softmax_cross_entropy(lg, target.cuda())
+    loss.backward()
+
+    tol = (1e-4, 1e-5) if dtype == torch.float32 else (2e-2, 2e-2)
+    assert abs(loss.item() - loss_ref.item()) < (1e-3 if dtype == torch.float32 else 5e-2)
+    _close(lg.grad, ref_l.grad, *tol, "xent dlogits")
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("channels_last", [False, True])
+def test_nhwc_pack(channels_last):
+    from tensorflowonspark_amd.ops.modules import nhwc_pack
+    torch.manual_seed(3)
+    x = torch.randint(0, 256, (4, 23, 23, 3), dtype=torch.uint8, device="cuda")
+    mean = torch.tensor([0.2, 0.3, 0.4], device="cuda")
+    std = torch.tensor([0.5, 0.6, 0.7], device="cuda")
+    out = nhwc_pack(x, mean, std, out_dtype=torch.bfloat16,
+                    channels_last=channels_last)
+    ref = nhwc_pack(x.cpu(), mean.cpu(), std.cpu(), out_dtype=torch.float32)
+    assert out.shape == ref.shape
+    if channels_last:
+        assert out.is_contiguous(memory_format=torch.channels_last)
+    _close(out, ref, 1e-2, 1e-2, "nhwc_pack")
+
+
+@gpu
+@requires_gpu
+def test_sgd_step():
+    ext = _ext()
+    torch.manual_seed(4)
+    n = 10007
+    p = torch.randn(n, device="cuda")
+    g = torch.randn(n, device="cuda")
+    m = torch.randn(n, device="cuda")
+    p_ref, g_ref, m_ref = p.clone(), g.clone(), m.clone()
+
+    ext.sgd_step(p, g, m, 0.1, 0.9, 1e-4, False)
+
+    grad = g_ref + 1e-4 * p_ref
+    m_exp = 0.9 * m_ref + grad
+    p_exp = p_ref - 0.1 * m_exp
+    _close(p, p_exp, 1e-5, 1e-6, "sgd p")
+    _close(m, m_exp, 1e-5, 1e-6, "sgd m")
+
+
+@gpu
+@requires_gpu
+def test_mfma_cd_layout():
+    """A=I-style probe with asymmetric B: verify the documented C/D mapping
+    col=lane&15, row=(lane>>4)*4+reg for mfma_f32_16x16x32_bf16."""
+    ext = _ext()
+    # build lane fragments for A[m,k]=delta(m,k-0)? simpler: random A,B 16x32
+    torch.manual_seed(5)
+    A = torch.randn(16, 32).bfloat16()
+    B = torch.randn(16, 32).bfloat16()   # "B^T" operand: B[n, k]
+    # assumed operand layout: lane l holds X[l&15][(l>>4)*8 + j]
+    a_l = torch.empty(64, 8, dtype=torch.int16)
+    b_l = torch.empty(64, 8, dtype=torch.int16)
+    for l in range(64):
+        r = l & 15
+        k0 = (l >> 4) * 8
+        a_l[l] = A[r, k0:k0 + 8].view(torch.int16)
+        b_l[l] = B[r, k0:k0 + 8].view(torch.int16)
+    c = ext.mfma_probe(a_l.cuda(), b_l.cuda()).cpu()
+    C_ref = A.float() @ B.float().t()   # [m, n]
+    C_got = torch.empty(16, 16)
+    for l in range(64):
+        for r in range(4):
+            C_got[(l >> 4) * 4 + r, l & 15] = c[l, r]
+    _close(C_got, C_ref, 1e-2, 5e-2, "mfma C/D layout")
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("mnk", [(128, 128, 128), (256, 512, 1024),
+                                 (100, 130, 96), (64, 64, 32)])
+def test_gemm_bt(mnk):
+    ext = _ext()
+    torch.manual_seed(6)
+    M, N, K = mnk
+    a = (torch.randn(M, K) / 8).bfloat16().cuda()
+    b = (torch.randn(N, K) / 8).bfloat16().cuda()
+    c = ext.gemm_bt(a, b)
+    ref = a.float().cpu() @ b.float().cpu().t()
+    _close(c, ref, 2e-2, 2e-2, "gemm_bt {}".format(mnk))
+
+
+@gpu
+@requires_gpu
+def test_resnet50_train_step():
+    """One full fwd+bwd+opt step of the flagship model on GPU."""
+    import __graft_entry__
+    __graft_entry__.smoke()
