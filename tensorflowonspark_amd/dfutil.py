@@ -1,0 +1,159 @@
+"""TFRecord <-> DataFrame conversion (parity: reference ``dfutil.py``).
+
+Works on both real Spark DataFrames and the local :class:`LocalDataFrame`.
+The TFRecord layer is the native codec in ``tfrecord.py`` (no tensorflow-hadoop
+jar, no TF protobuf dependency).
+
+Dtype mapping parity (reference ``dfutil.py:96-131``): float/double ->
+FloatList, bool/ints -> Int64List, binary/string -> BytesList, arrays thereof.
+Schema inference parity (reference ``dfutil.py:134-168``): int64 -> bigint,
+float -> double, bytes -> string unless hinted in ``binary_features``;
+multi-value features -> arrays.
+"""
+
+import glob
+import logging
+import os
+
+from . import tfrecord
+
+logger = logging.getLogger(__name__)
+
+loadedDF = {}  # DataFrame provenance: df -> source dir (reference dfutil.py:18-26)
+
+
+def isLoadedDF(df):
+    """True if ``df`` was produced by :func:`loadTFRecords`."""
+    return id(df) in loadedDF
+
+
+def toTFExample(dtypes):
+    """Returns an iterator closure converting DataFrame rows (in the column
+    order of ``dtypes``) to serialized Example bytes."""
+    cols = [name for name, _t in dtypes]
+    types = [t for _n, t in dtypes]
+
+    def _convert(iterator):
+        for row in iterator:
+            feats = {}
+            for name, t, v in zip(cols, types, row):
+                if v is None:
+                    v = []
+                if t in ("float", "double") or t.startswith("array<float>") \
+                        or t.startswith("array<double>"):
+                    feats[name] = [float(x) for x in (v if isinstance(v, (list, tuple)) else [v])]
+                elif t in ("boolean", "tinyint", "smallint", "int", "bigint", "long") \
+                        or t.startswith("array<int") or t.startswith("array<bigint") \
+                        or t.startswith("array<long") or t.startswith("array<boolean"):
+                    feats[name] = [int(x) for x in (v if isinstance(v, (list, tuple)) else [v])]
+                elif t in ("binary", "string") or t.startswith("array<binary") \
+                        or t.startswith("array<string"):
+                    vals = v if isinstance(v, (list, tuple)) else [v]
+                    feats[name] = [x if isinstance(x, (bytes, bytearray))
+                                   else str(x).encode("utf-8") for x in vals]
+                else:
+                    raise TypeError("unsupported dtype {} for column {}".format(t, name))
+            yield tfrecord.encode_example(feats)
+
+    return _convert
+
+
+def infer_schema(example, binary_features=None):
+    """Infer (name, dtype) schema from one decoded Example
+    ({name: (kind, values)})."""
+    binary_features = set(binary_features or [])
+    schema = []
+    for name in sorted(example.keys()):
+        kind, values = example[name]
+        if kind == "int64":
+            base = "bigint"
+        elif kind == "float":
+            base = "double"
+        else:
+            base = "binary" if name in binary_features else "string"
+        if len(values) > 1:
+            schema.append((name, "array<{}>".format(base)))
+        else:
+            schema.append((name, base))
+    return schema
+
+
+def fromTFExample(record, binary_features=None, schema=None):
+    """Serialized Example -> row tuple (columns sorted by name)."""
+    binary_features = set(binary_features or [])
+    ex = tfrecord.decode_example(record)
+    names = [n for n, _t in schema] if schema else sorted(ex.keys())
+    row = []
+    for name in names:
+        if name not in ex:
+            row.append(None)
+            continue
+        kind, values = ex[name]
+        if kind == "bytes" and name not in binary_features:
+            values = [v.decode("utf-8", errors="replace") for v in values]
+        if len(values) == 0:
+            row.append(None)
+        elif len(values) == 1 and not (schema and
+                                       dict(schema)[name].startswith("array")):
+            row.append(values[0])
+        else:
+            row.append(list(values))
+    return tuple(row)
+
+
+def saveAsTFRecords(df, output_dir):
+    """Save a DataFrame as TFRecord part files under ``output_dir``."""
+    dtypes = df.dtypes
+    convert = toTFExample(dtypes)
+    os.makedirs(output_dir, exist_ok=True)
+
+    def _write(idx, iterator):
+        path = os.path.join(output_dir, "part-r-{:05d}".format(idx))
+        n = 0
+        with tfrecord.TFRecordWriter(path) as w:
+            for rec in convert(iterator):
+                w.write(rec)
+                n += 1
+        return [n]
+
+    rdd = df.rdd
+    if hasattr(rdd, "mapPartitionsWithIndex"):
+        counts = rdd.mapPartitionsWithIndex(_write).collect()
+        logger.info("wrote %s records to %s", sum(counts), output_dir)
+    else:  # pragma: no cover
+        with tfrecord.TFRecordWriter(os.path.join(output_dir, "part-r-00000")) as w:
+            for rec in convert(iter(df.collect())):
+                w.write(rec)
+
+
+def loadTFRecords(sc, input_dir, binary_features=None):
+    """Load TFRecord files under ``input_dir`` as a DataFrame with inferred
+    schema; records provenance in ``loadedDF``."""
+    files = sorted(f for f in glob.glob(os.path.join(input_dir, "part-*"))
+                   if os.path.isfile(f))
+    if not files:
+        files = sorted(f for f in glob.glob(os.path.join(input_dir, "*"))
+                       if os.path.isfile(f) and not os.path.basename(f).startswith("_"))
+    if not files:
+        raise FileNotFoundError("no TFRecord files under " + input_dir)
+
+    first = next(tfrecord.tfrecord_iterator(files[0]))
+    schema = infer_schema(tfrecord.decode_example(first), binary_features)
+
+    def _read(it):
+        for path in it:
+            for rec in tfrecord.tfrecord_iterator(path):
+                yield fromTFExample(rec, binary_features, schema)
+
+    rows = sc.parallelize(files, min(len(files), sc.defaultParallelism)) \
+        .mapPartitions(_read).collect()
+    if hasattr(sc, "createDataFrame"):
+        df = sc.createDataFrame(rows, [n for n, _ in schema],
+                                [t for _, t in schema])
+    else:  # pragma: no cover - real pyspark path
+        from pyspark.sql import SparkSession
+        spark = SparkSession.builder.getOrCreate()
+        df = spark.createDataFrame(rows, " ".join(
+            "{}:{}".format(n, t) for n, t in schema))
+    loadedDF[id(df)] = input_dir
+    return df

@@ -42,6 +42,10 @@ logger = logging.getLogger(__name__)
 _STOP = "__stop__"
 
 
+def _index_wrap(fn, idx, it):
+    return fn(idx, it)
+
+
 def _executor_main(exec_id, workdir, task_q, result_q, env):
     import cloudpickle
     os.makedirs(workdir, exist_ok=True)
@@ -82,6 +86,13 @@ class LocalRDD:
     def mapPartitions(self, fn):
         return _LazyRDD(self.sc, self.partitions, fn)
 
+    def mapPartitionsWithIndex(self, fn):
+        return _LazyRDD(self.sc, self.partitions, fn, with_index=True)
+
+    def map(self, fn):
+        return _LazyRDD(self.sc, self.partitions,
+                        lambda it: (fn(x) for x in it))
+
     def collect(self):
         return [x for p in self.partitions for x in p]
 
@@ -92,10 +103,11 @@ class LocalRDD:
 class _LazyRDD:
     """mapPartitions result; executes on .collect()/.foreachPartition()."""
 
-    def __init__(self, sc, partitions, fn):
+    def __init__(self, sc, partitions, fn, with_index=False):
         self.sc = sc
         self.partitions = partitions
         self.fn = fn
+        self.with_index = with_index
 
     def getNumPartitions(self):
         return len(self.partitions)
@@ -108,7 +120,8 @@ class _LazyRDD:
         return _LazyRDD(self.sc, self.partitions, chained)
 
     def collect(self):
-        results = self.sc._run_job(self.partitions, self.fn, collect=True)
+        results = self.sc._run_job(self.partitions, self.fn, collect=True,
+                                   with_index=self.with_index)
         return [x for p in results for x in (p or [])]
 
     def foreachPartition(self, fn2):
@@ -120,17 +133,81 @@ class _LazyRDD:
         self.sc._run_job(self.partitions, chained, collect=False)
 
 
+def _infer_dtype(v):
+    if isinstance(v, bool):
+        return "boolean"
+    if isinstance(v, int):
+        return "bigint"
+    if isinstance(v, float):
+        return "double"
+    if isinstance(v, str):
+        return "string"
+    if isinstance(v, (bytes, bytearray)):
+        return "binary"
+    if isinstance(v, (list, tuple)):
+        inner = _infer_dtype(v[0]) if v else "double"
+        return "array<{}>".format(inner)
+    return "string"
+
+
+class LocalDataFrame:
+    """Minimal Spark-SQL-DataFrame stand-in: named, typed columns over rows.
+
+    Supports the subset the pipeline/dfutil layers use: ``columns``,
+    ``dtypes``, ``select``, ``.rdd``, ``collect``, ``count``.
+    """
+
+    def __init__(self, sc, rows, columns, dtypes=None):
+        self.sc = sc
+        self._rows = [tuple(r) for r in rows]
+        self.columns = list(columns)
+        if dtypes is None:
+            first = self._rows[0] if self._rows else tuple("" for _ in columns)
+            dtypes = [_infer_dtype(v) for v in first]
+        self.dtypes = list(zip(self.columns, dtypes))
+
+    def select(self, *cols):
+        if len(cols) == 1 and isinstance(cols[0], (list, tuple)):
+            cols = list(cols[0])
+        idx = [self.columns.index(c) for c in cols]
+        rows = [tuple(r[i] for i in idx) for r in self._rows]
+        dt = [self.dtypes[i][1] for i in idx]
+        return LocalDataFrame(self.sc, rows, list(cols), dt)
+
+    @property
+    def rdd(self):
+        return self.sc.parallelize(self._rows, self.sc.defaultParallelism)
+
+    def collect(self):
+        return list(self._rows)
+
+    def count(self):
+        return len(self._rows)
+
+    def show(self, n=20):
+        print(self.columns)
+        for r in self._rows[:n]:
+            print(r)
+
+
 class _Job:
-    def __init__(self, job_id, partitions, fn, collect):
+    def __init__(self, job_id, partitions, fn, collect, with_index=False):
         self.job_id = job_id
         self.partitions = partitions
         self.fn = fn
         self.collect = collect
+        self.with_index = with_index
         self.pending = list(range(len(partitions)))
         self.results = [None] * len(partitions)
         self.done = 0
         self.error = None
         self.event = threading.Event()
+
+    def task_fn(self, pid):
+        import functools
+        if self.with_index:
+            return functools.partial(_index_wrap, self.fn, pid)
+        return self.fn
 
 
 class LocalSparkContext:
@@ -191,6 +268,9 @@ class LocalSparkContext:
             parts.extend(r.partitions)
         return LocalRDD(self, parts)
 
+    def createDataFrame(self, data, columns, dtypes=None):
+        return LocalDataFrame(self, data, columns, dtypes)
+
     def cancelAllJobs(self):
         pass
 
@@ -212,11 +292,12 @@ class LocalSparkContext:
 
     # -- scheduler ------------------------------------------------------------
 
-    def _run_job(self, partitions, fn, collect, pin=False, timeout=None):
+    def _run_job(self, partitions, fn, collect, pin=False, timeout=None,
+                 with_index=False):
         """Submit one task per partition; block until all complete or one fails."""
         if self._stopped:
             raise RuntimeError("context is stopped")
-        job = _Job(uuid.uuid4().hex, partitions, fn, collect)
+        job = _Job(uuid.uuid4().hex, partitions, fn, collect, with_index)
         with self._lock:
             self._jobs[job.job_id] = job
             if pin:
@@ -225,7 +306,8 @@ class LocalSparkContext:
                         len(partitions), self.num_executors)
                 for pid in list(job.pending):
                     self._task_qs[pid].put(cloudpickle.dumps(
-                        (job.job_id, pid, fn, partitions[pid], collect)))
+                        (job.job_id, pid, job.task_fn(pid), partitions[pid],
+                         collect)))
                     self._busy[pid] += 1
                 job.pending = []
             else:
@@ -279,7 +361,8 @@ class LocalSparkContext:
             eid = idle.pop()
             pid = job.pending.pop(0)
             self._task_qs[eid].put(cloudpickle.dumps(
-                (job.job_id, pid, job.fn, job.partitions[pid], job.collect)))
+                (job.job_id, pid, job.task_fn(pid), job.partitions[pid],
+                 job.collect)))
             self._busy[eid] += 1
             if not job.pending:
                 self._job_order.pop(0)
