@@ -9,15 +9,15 @@ extern "C" {
 void tfosr_bn_stats(const void*, int, int, float*, float*, int, int, long, hipStream_t);
 void tfosr_bn_finalize(const float*, const float*, float*, float*, float*, float*,
                        long, int, float, float, hipStream_t);
-void tfosr_bn_relu_apply(const void*, void*, const float*, const float*,
-                         const float*, const float*, int, int, long, int, long,
-                         hipStream_t);
-void tfosr_bn_bwd_stats(const void*, const void*, const void*, const float*,
-                        const float*, float*, float*, int, int, int, int, long,
-                        hipStream_t);
+void tfosr_bn_apply(const void*, const void*, void*, const float*, const float*,
+                    const float*, const float*, int, int, int, long, int, long,
+                    hipStream_t);
+void tfosr_bn_bwd_stats(const void*, const void*, const void*, void*,
+                        const float*, const float*, float*, float*, int, int,
+                        int, int, int, long, hipStream_t);
 void tfosr_bn_bwd_dx(const void*, const void*, const void*, const float*,
                      const float*, const float*, const float*, const float*,
-                     void*, int, int, long, int, long, hipStream_t);
+                     void*, int, int, int, long, int, long, hipStream_t);
 void tfosr_xent_fwd(const void*, const long*, float*, float*, int, int, int,
                     hipStream_t);
 void tfosr_xent_bwd(const void*, const float*, const long*, const float*, void*,
@@ -64,11 +64,17 @@ int dtype_flag(const at::Tensor& x) {
   return 0;
 }
 
-std::vector<at::Tensor> bn_relu_fwd_train(at::Tensor x, at::Tensor w, at::Tensor b,
-                                          at::Tensor rm, at::Tensor rv,
-                                          double momentum, double eps) {
+std::vector<at::Tensor> bn_fwd_train(at::Tensor x, c10::optional<at::Tensor> res,
+                                     at::Tensor w, at::Tensor b, at::Tensor rm,
+                                     at::Tensor rv, double momentum, double eps,
+                                     bool relu) {
   auto l = bn_layout(x);
   int bf = dtype_flag(x);
+  const void* res_ptr = nullptr;
+  if (res.has_value()) {
+    TORCH_CHECK(res->sizes() == x.sizes() && res->scalar_type() == x.scalar_type());
+    res_ptr = res->data_ptr();
+  }
   auto opts = x.options().dtype(at::kFloat);
   auto wsum = at::zeros({l.C}, opts);
   auto wsq = at::zeros({l.C}, opts);
@@ -83,49 +89,61 @@ std::vector<at::Tensor> bn_relu_fwd_train(at::Tensor x, at::Tensor w, at::Tensor
                     save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                     rm.data_ptr<float>(), rv.data_ptr<float>(), M, l.C,
                     (float)momentum, (float)eps, s);
-  tfosr_bn_relu_apply(x.data_ptr(), y.data_ptr(), save_mean.data_ptr<float>(),
-                      save_rstd.data_ptr<float>(), w.data_ptr<float>(),
-                      b.data_ptr<float>(), bf, l.nhwc, M * l.C, l.C, l.HW, s);
+  tfosr_bn_apply(x.data_ptr(), res_ptr, y.data_ptr(), save_mean.data_ptr<float>(),
+                 save_rstd.data_ptr<float>(), w.data_ptr<float>(),
+                 b.data_ptr<float>(), bf, l.nhwc, relu, M * l.C, l.C, l.HW, s);
   return {y, save_mean, save_rstd};
 }
 
-at::Tensor bn_relu_fwd_eval(at::Tensor x, at::Tensor w, at::Tensor b,
-                            at::Tensor rm, at::Tensor rv, double eps) {
+at::Tensor bn_fwd_eval(at::Tensor x, c10::optional<at::Tensor> res, at::Tensor w,
+                       at::Tensor b, at::Tensor rm, at::Tensor rv, double eps,
+                       bool relu) {
   auto l = bn_layout(x);
   int bf = dtype_flag(x);
+  const void* res_ptr = res.has_value() ? res->data_ptr() : nullptr;
   auto rstd = at::rsqrt(rv + eps);
   auto y = at::empty_like(x);
-  tfosr_bn_relu_apply(x.data_ptr(), y.data_ptr(), rm.data_ptr<float>(),
-                      rstd.data_ptr<float>(), w.data_ptr<float>(),
-                      b.data_ptr<float>(), bf, l.nhwc,
-                      (long)l.N * l.HW * l.C, l.C, l.HW, cur_stream());
+  tfosr_bn_apply(x.data_ptr(), res_ptr, y.data_ptr(), rm.data_ptr<float>(),
+                 rstd.data_ptr<float>(), w.data_ptr<float>(), b.data_ptr<float>(),
+                 bf, l.nhwc, relu, (long)l.N * l.HW * l.C, l.C, l.HW,
+                 cur_stream());
   return y;
 }
 
-std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
-                                    at::Tensor w, at::Tensor save_mean,
-                                    at::Tensor save_rstd) {
+std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
+                               at::Tensor w, at::Tensor save_mean,
+                               at::Tensor save_rstd, bool relu, bool grad_res) {
   auto l = bn_layout(x);
   int bf = dtype_flag(x);
-  if (l.nhwc) {
-    dy = dy.contiguous(at::MemoryFormat::ChannelsLast);
-  } else {
-    dy = dy.contiguous();
-  }
+  dy = l.nhwc ? dy.contiguous(at::MemoryFormat::ChannelsLast) : dy.contiguous();
   auto opts = x.options().dtype(at::kFloat);
   auto dg = at::zeros({l.C}, opts);
   auto db = at::zeros({l.C}, opts);
   auto dx = at::empty_like(x);
+  at::Tensor gout;
+  void* gout_ptr = nullptr;
+  if (grad_res) {
+    gout = at::empty_like(x);
+    gout_ptr = gout.data_ptr();
+  }
   auto s = cur_stream();
-  tfosr_bn_bwd_stats(x.data_ptr(), dy.data_ptr(), y.data_ptr(),
+  tfosr_bn_bwd_stats(x.data_ptr(), dy.data_ptr(), y.data_ptr(), gout_ptr,
                      save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                      dg.data_ptr<float>(), db.data_ptr<float>(), bf, l.nhwc,
-                     l.N, l.C, l.HW, s);
+                     relu, l.N, l.C, l.HW, s);
+  long total = (long)l.N * l.HW * l.C;
+  if (grad_res) {
+    // gout already carries the gated gradient: skip re-gating (and the y read)
+    tfosr_bn_bwd_dx(x.data_ptr(), gout.data_ptr(), y.data_ptr(),
+                    save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
+                    w.data_ptr<float>(), dg.data_ptr<float>(), db.data_ptr<float>(),
+                    dx.data_ptr(), bf, l.nhwc, /*relu=*/0, total, l.C, l.HW, s);
+    return {dx, dg, db, gout};
+  }
   tfosr_bn_bwd_dx(x.data_ptr(), dy.data_ptr(), y.data_ptr(),
                   save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                   w.data_ptr<float>(), dg.data_ptr<float>(), db.data_ptr<float>(),
-                  dx.data_ptr(), bf, l.nhwc, (long)l.N * l.HW * l.C, l.C, l.HW, s);
-  // dgamma/dbeta are the reduced sums (save_* already folded into dg)
+                  dx.data_ptr(), bf, l.nhwc, relu, total, l.C, l.HW, s);
   return {dx, dg, db};
 }
 
@@ -220,9 +238,9 @@ at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("bn_relu_fwd_train", &bn_relu_fwd_train);
-  m.def("bn_relu_fwd_eval", &bn_relu_fwd_eval);
-  m.def("bn_relu_bwd", &bn_relu_bwd);
+  m.def("bn_fwd_train", &bn_fwd_train);
+  m.def("bn_fwd_eval", &bn_fwd_eval);
+  m.def("bn_bwd", &bn_bwd);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("nhwc_pack", &nhwc_pack);

@@ -1,75 +1,281 @@
-// Fused BatchNorm2d + ReLU for gfx950 — training fwd (stats + normalize+relu),
-// eval fwd, and fused backward (reductions + dx), NCHW and NHWC layouts,
-// fp32 or bf16 activations with fp32 statistics.
+// Fused BatchNorm2d (+residual add)(+ReLU) for gfx950 — training fwd/bwd and
+// eval fwd, NHWC (channels_last) fast path + generic fallbacks, fp32/bf16
+// activations with fp32 statistics.
 //
-// These ops are HBM-bandwidth-bound on MI355X (8 TB/s peak, ~6.3 achievable):
-// the win vs separate BN/ReLU kernels is eliminating whole-tensor round trips —
-// fwd reads x twice + writes y once (3 passes total vs 5 unfused), bwd reads
-// x,dy twice + writes dx once (5 vs 8). All loads vectorized 16 B/lane.
+// These ops are HBM-bound on MI355X; the design rules applied here:
+//  * 16 B/lane vector access everywhere (8 bf16 / 4 f32 per thread) — hipcc
+//    does not auto-vectorize bf16 loads.
+//  * all hot-loop index math in uint32 (64-bit div/mod is ~40 VALU cycles and
+//    made the first version 10x off roofline); launchers fall back to the
+//    generic kernels when a tensor exceeds 2^31 elements.
+//  * NHWC reductions: thread t owns the 8 fixed channels (t*8..t*8+7) mod C
+//    (valid when C | 2048 — every ResNet/UNet width), accumulating in
+//    registers over a grid-stride loop, one atomicAdd per channel at the end.
+//  * fusion: the residual add and ReLU fold into the normalize pass (fwd) and
+//    the dgamma/dbeta pass emits the gated upstream gradient (bwd) so the
+//    elementwise add/relu/relu' kernels and their whole-tensor round trips
+//    disappear.
 #include "tfosr_common.h"
 
+typedef unsigned int u32;
+
 // ---------------------------------------------------------------------------
-// Stage 1: per-channel sum / sum-of-squares partials
+// load/store helpers: VEC elements of T as one 16 B transaction
 // ---------------------------------------------------------------------------
 
-// NHWC: x viewed as [M, C]; adjacent lanes read adjacent channels (coalesced).
 template <typename T>
-__global__ void stats_nhwc_kernel(const T* __restrict__ x, float* __restrict__ wsum,
-                                  float* __restrict__ wsq, long M, int C) {
-  const long rows_per_blk = (M + gridDim.x - 1) / gridDim.x;
-  const long r0 = blockIdx.x * rows_per_blk;
-  const long r1 = min(M, r0 + rows_per_blk);
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    float s = 0.f, q = 0.f;
-    for (long r = r0; r < r1; ++r) {
-      float v = (float)x[r * C + c];
-      s += v;
-      q += v * v;
+struct VecIO;
+
+template <>
+struct VecIO<bf16_t> {
+  static constexpr int V = 8;
+  using vec_t = s8v;
+  __device__ static void load(const bf16_t* p, float* out) {
+    vec_t v = *(const vec_t*)p;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      union { short s; bf16_t b; } u;
+      u.s = v[j];
+      out[j] = (float)u.b;
     }
-    atomicAdd(&wsum[c], s);
-    atomicAdd(&wsq[c], q);
+  }
+  __device__ static void store(bf16_t* p, const float* in) {
+    vec_t v;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      union { short s; bf16_t b; } u;
+      u.b = (bf16_t)in[j];
+      v[j] = u.s;
+    }
+    *(vec_t*)p = v;
+  }
+};
+
+template <>
+struct VecIO<float> {
+  static constexpr int V = 4;
+  using vec_t = f4v;
+  __device__ static void load(const float* p, float* out) {
+    f4v v = *(const f4v*)p;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) out[j] = v[j];
+  }
+  __device__ static void store(float* p, const float* in) {
+    f4v v;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) v[j] = in[j];
+    *(f4v*)p = v;
+  }
+};
+
+// ---------------------------------------------------------------------------
+// NHWC fast path — requires C % V == 0 and 2048 % C == 0 (fixed per-thread
+// channel slots) and total < 2^31
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void stats_nhwc_fast(const T* __restrict__ x, float* __restrict__ wsum,
+                                float* __restrict__ wsq, u32 nvec, u32 C) {
+  constexpr int V = VecIO<T>::V;
+  const u32 c0 = ((u32)threadIdx.x * V) % C;
+  float s[V], q[V], v[V];
+  #pragma unroll
+  for (int j = 0; j < V; ++j) { s[j] = 0.f; q[j] = 0.f; }
+  const u32 stride = gridDim.x * blockDim.x;
+  for (u32 i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec; i += stride) {
+    VecIO<T>::load(x + (size_t)i * V, v);
+    #pragma unroll
+    for (int j = 0; j < V; ++j) { s[j] += v[j]; q[j] += v[j] * v[j]; }
+  }
+  #pragma unroll
+  for (int j = 0; j < V; ++j) {
+    atomicAdd(&wsum[c0 + j], s[j]);
+    atomicAdd(&wsq[c0 + j], q[j]);
   }
 }
 
-// NCHW: one (channel, split) pair per block; vectorized 8-wide over HW.
-template <typename T, int VEC>
-__global__ void stats_nchw_kernel(const T* __restrict__ x, float* __restrict__ wsum,
-                                  float* __restrict__ wsq, int N, int C, long HW) {
-  __shared__ float scratch[8];
-  const int c = blockIdx.x % C;
-  const int split = blockIdx.x / C;
-  const int nsplit = gridDim.x / C;
-  float s = 0.f, q = 0.f;
-  const long chunk = (HW / VEC + nsplit - 1) / nsplit;
-  const long v0 = split * chunk, v1 = min(HW / VEC, v0 + chunk);
-  for (int n = 0; n < N; ++n) {
-    const T* plane = x + ((long)n * C + c) * HW;
-    for (long i = v0 + threadIdx.x; i < v1; i += blockDim.x) {
-      #pragma unroll
-      for (int j = 0; j < VEC; ++j) {
-        float v = (float)plane[i * VEC + j];
-        s += v;
-        q += v * v;
-      }
-    }
-    if (split == 0) {  // scalar tail
-      for (long i = (HW / VEC) * VEC + threadIdx.x; i < HW; i += blockDim.x) {
-        float v = (float)plane[i];
-        s += v;
-        q += v * v;
-      }
-    }
+// normalize (+add residual)(+relu); optionally store the pre-activation sum
+template <typename T, bool RELU, bool ADD>
+__global__ void bn_apply_fast(const T* __restrict__ x, const T* __restrict__ res,
+                              T* __restrict__ y,
+                              const float* __restrict__ mean,
+                              const float* __restrict__ rstd,
+                              const float* __restrict__ w,
+                              const float* __restrict__ b, u32 nvec, u32 C) {
+  constexpr int V = VecIO<T>::V;
+  const u32 c0 = ((u32)threadIdx.x * V) % C;
+  float sc[V], sh[V], v[V], r[V];
+  #pragma unroll
+  for (int j = 0; j < V; ++j) {
+    sc[j] = w[c0 + j] * rstd[c0 + j];
+    sh[j] = b[c0 + j] - mean[c0 + j] * sc[j];
   }
-  s = block_sum<256>(s, scratch);
-  __syncthreads();
-  q = block_sum<256>(q, scratch);
-  if (threadIdx.x == 0) {
-    atomicAdd(&wsum[c], s);
-    atomicAdd(&wsq[c], q);
+  const u32 stride = gridDim.x * blockDim.x;
+  for (u32 i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec; i += stride) {
+    VecIO<T>::load(x + (size_t)i * V, v);
+    if (ADD) VecIO<T>::load(res + (size_t)i * V, r);
+    #pragma unroll
+    for (int j = 0; j < V; ++j) {
+      float o = v[j] * sc[j] + sh[j];
+      if (ADD) o += r[j];
+      if (RELU) o = fmaxf(o, 0.f);
+      v[j] = o;
+    }
+    VecIO<T>::store(y + (size_t)i * V, v);
   }
 }
 
-// Stage 2: finalize mean/rstd + update running stats. C threads total.
+// backward reductions: g = dy * gate(y); dbeta += g; dgamma += g * xhat;
+// optionally write g out (the residual branch gradient for the ADD variant)
+template <typename T, bool RELU, bool WRITE_G>
+__global__ void bwd_stats_fast(const T* __restrict__ x, const T* __restrict__ dy,
+                               const T* __restrict__ y, T* __restrict__ gout,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ rstd,
+                               float* __restrict__ dg, float* __restrict__ db,
+                               u32 nvec, u32 C) {
+  constexpr int V = VecIO<T>::V;
+  const u32 c0 = ((u32)threadIdx.x * V) % C;
+  float sg[V], sb[V], xv[V], dv[V], yv[V];
+  float mu[V], rs[V];
+  #pragma unroll
+  for (int j = 0; j < V; ++j) {
+    sg[j] = 0.f; sb[j] = 0.f;
+    mu[j] = mean[c0 + j]; rs[j] = rstd[c0 + j];
+  }
+  const u32 stride = gridDim.x * blockDim.x;
+  for (u32 i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec; i += stride) {
+    VecIO<T>::load(x + (size_t)i * V, xv);
+    VecIO<T>::load(dy + (size_t)i * V, dv);
+    if (RELU) VecIO<T>::load(y + (size_t)i * V, yv);
+    #pragma unroll
+    for (int j = 0; j < V; ++j) {
+      float g = RELU ? (yv[j] > 0.f ? dv[j] : 0.f) : dv[j];
+      sb[j] += g;
+      sg[j] += g * (xv[j] - mu[j]) * rs[j];
+      dv[j] = g;
+    }
+    if (WRITE_G) VecIO<T>::store(gout + (size_t)i * V, dv);
+  }
+  #pragma unroll
+  for (int j = 0; j < V; ++j) {
+    atomicAdd(&dg[c0 + j], sg[j]);
+    atomicAdd(&db[c0 + j], sb[j]);
+  }
+}
+
+// dx = w*rstd * (g - db/M - xhat * dg/M); g recomputed from (dy, y) or read
+// from the gated gradient written by bwd_stats (GATED=false).
+template <typename T, bool RELU>
+__global__ void bwd_dx_fast(const T* __restrict__ x, const T* __restrict__ dy,
+                            const T* __restrict__ y, T* __restrict__ dx,
+                            const float* __restrict__ mean,
+                            const float* __restrict__ rstd,
+                            const float* __restrict__ w,
+                            const float* __restrict__ dg,
+                            const float* __restrict__ db,
+                            u32 nvec, u32 C, float invM) {
+  constexpr int V = VecIO<T>::V;
+  const u32 c0 = ((u32)threadIdx.x * V) % C;
+  float sc[V], mu[V], rs[V], dgm[V], dbm[V], xv[V], dv[V], yv[V];
+  #pragma unroll
+  for (int j = 0; j < V; ++j) {
+    mu[j] = mean[c0 + j]; rs[j] = rstd[c0 + j];
+    sc[j] = w[c0 + j] * rs[j];
+    dgm[j] = dg[c0 + j] * invM;
+    dbm[j] = db[c0 + j] * invM;
+  }
+  const u32 stride = gridDim.x * blockDim.x;
+  for (u32 i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec; i += stride) {
+    VecIO<T>::load(x + (size_t)i * V, xv);
+    VecIO<T>::load(dy + (size_t)i * V, dv);
+    if (RELU) VecIO<T>::load(y + (size_t)i * V, yv);
+    #pragma unroll
+    for (int j = 0; j < V; ++j) {
+      float g = RELU ? (yv[j] > 0.f ? dv[j] : 0.f) : dv[j];
+      float xhat = (xv[j] - mu[j]) * rs[j];
+      dv[j] = sc[j] * (g - dbm[j] - xhat * dgm[j]);
+    }
+    VecIO<T>::store(dx + (size_t)i * V, dv);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Generic fallbacks (any layout/shape): scalar, 64-bit safe
+// ---------------------------------------------------------------------------
+
+template <typename T, bool NHWC>
+__global__ void stats_generic(const T* __restrict__ x, float* __restrict__ wsum,
+                              float* __restrict__ wsq, long total, int C, long HW) {
+  __shared__ float red[64 * 2];
+  // per-channel loop with block reduction would be slow for big C; use
+  // atomic per element strided — acceptable: this path only serves odd shapes
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    int c = NHWC ? (int)(idx % C) : (int)((idx / HW) % C);
+    float v = (float)x[idx];
+    atomicAdd(&wsum[c], v);
+    atomicAdd(&wsq[c], v * v);
+  }
+  (void)red;
+}
+
+template <typename T, bool NHWC, bool RELU, bool ADD>
+__global__ void bn_apply_generic(const T* __restrict__ x, const T* __restrict__ res,
+                                 T* __restrict__ y,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ rstd,
+                                 const float* __restrict__ w,
+                                 const float* __restrict__ b,
+                                 long total, int C, long HW) {
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    int c = NHWC ? (int)(idx % C) : (int)((idx / HW) % C);
+    float o = ((float)x[idx] - mean[c]) * rstd[c] * w[c] + b[c];
+    if (ADD) o += (float)res[idx];
+    if (RELU) o = fmaxf(o, 0.f);
+    y[idx] = (T)o;
+  }
+}
+
+template <typename T, bool NHWC, bool RELU, bool WRITE_G>
+__global__ void bwd_stats_generic(const T* __restrict__ x, const T* __restrict__ dy,
+                                  const T* __restrict__ y, T* __restrict__ gout,
+                                  const float* __restrict__ mean,
+                                  const float* __restrict__ rstd,
+                                  float* __restrict__ dg, float* __restrict__ db,
+                                  long total, int C, long HW) {
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    int c = NHWC ? (int)(idx % C) : (int)((idx / HW) % C);
+    float g = (float)dy[idx];
+    if (RELU) g = ((float)y[idx] > 0.f) ? g : 0.f;
+    atomicAdd(&db[c], g);
+    atomicAdd(&dg[c], g * ((float)x[idx] - mean[c]) * rstd[c]);
+    if (WRITE_G) gout[idx] = (T)g;
+  }
+}
+
+template <typename T, bool NHWC, bool RELU>
+__global__ void bwd_dx_generic(const T* __restrict__ x, const T* __restrict__ dy,
+                               const T* __restrict__ y, T* __restrict__ dx,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ rstd,
+                               const float* __restrict__ w,
+                               const float* __restrict__ dg,
+                               const float* __restrict__ db,
+                               long total, int C, long HW, float invM) {
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    int c = NHWC ? (int)(idx % C) : (int)((idx / HW) % C);
+    float g = (float)dy[idx];
+    if (RELU) g = ((float)y[idx] > 0.f) ? g : 0.f;
+    float xhat = ((float)x[idx] - mean[c]) * rstd[c];
+    dx[idx] = (T)(w[c] * rstd[c] * (g - db[c] * invM - xhat * dg[c] * invM));
+  }
+}
+
+// finalize: mean/rstd + running-stat update (C threads)
 __global__ void stats_finalize_kernel(const float* __restrict__ wsum,
                                       const float* __restrict__ wsq,
                                       float* __restrict__ save_mean,
@@ -91,200 +297,54 @@ __global__ void stats_finalize_kernel(const float* __restrict__ wsum,
 }
 
 // ---------------------------------------------------------------------------
-// Stage 3: normalize + ReLU elementwise (vectorized 8/lane)
+// extern "C" launchers — dispatch fast path when eligible
 // ---------------------------------------------------------------------------
 
-template <typename T, bool NHWC>
-__global__ void bn_relu_apply_kernel(const T* __restrict__ x, T* __restrict__ y,
-                                     const float* __restrict__ mean,
-                                     const float* __restrict__ rstd,
-                                     const float* __restrict__ w,
-                                     const float* __restrict__ b,
-                                     long total, int C, long HW) {
-  constexpr int V = 8;
-  const long nvec = total / V;
-  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
-       i += (long)gridDim.x * blockDim.x) {
-    const long base = i * V;
-    T vin[V], vout[V];
-    *(s8v*)vin = *(const s8v*)(x + base);  // 16 B vector load (T is 2 or 4 B -> use per-T below)
-    #pragma unroll
-    for (int j = 0; j < V; ++j) {
-      long idx = base + j;
-      int c = NHWC ? (int)(idx % C) : (int)((idx / HW) % C);
-      float v = (float)vin[j];
-      float r = (v - mean[c]) * rstd[c] * w[c] + b[c];
-      vout[j] = (T)fmaxf(r, 0.f);
-    }
-    *(s8v*)(y + base) = *(const s8v*)vout;
-  }
-  // tail
-  long tail0 = nvec * V;
-  for (long idx = tail0 + blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    int c = NHWC ? (int)(idx % C) : (int)((idx / HW) % C);
-    float v = (float)x[idx];
-    float r = (v - mean[c]) * rstd[c] * w[c] + b[c];
-    y[idx] = (T)fmaxf(r, 0.f);
-  }
+static inline bool fast_ok(int is_nhwc, long total, int C, int vec) {
+  return is_nhwc && total < (1L << 31) && C % vec == 0 && 2048 % C == 0;
 }
 
-// fp32 specialization needs 32B per 8 elems — split into two f4v ops
-template <bool NHWC>
-__global__ void bn_relu_apply_f32_kernel(const float* __restrict__ x,
-                                         float* __restrict__ y,
-                                         const float* __restrict__ mean,
-                                         const float* __restrict__ rstd,
-                                         const float* __restrict__ w,
-                                         const float* __restrict__ b,
-                                         long total, int C, long HW) {
-  constexpr int V = 4;
-  const long nvec = total / V;
-  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
-       i += (long)gridDim.x * blockDim.x) {
-    const long base = i * V;
-    f4v vin = *(const f4v*)(x + base);
-    f4v vout;
-    #pragma unroll
-    for (int j = 0; j < V; ++j) {
-      long idx = base + j;
-      int c = NHWC ? (int)(idx % C) : (int)((idx / HW) % C);
-      float r = (vin[j] - mean[c]) * rstd[c] * w[c] + b[c];
-      vout[j] = fmaxf(r, 0.f);
-    }
-    *(f4v*)(y + base) = vout;
-  }
-  long tail0 = nvec * V;
-  for (long idx = tail0 + blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    int c = NHWC ? (int)(idx % C) : (int)((idx / HW) % C);
-    float r = (x[idx] - mean[c]) * rstd[c] * w[c] + b[c];
-    y[idx] = fmaxf(r, 0.f);
-  }
+static inline int fast_grid(long nvec) {
+  long g = (nvec + 255) / 256;
+  if (g > TFOSR_MAX_GRID) g = TFOSR_MAX_GRID;
+  // keep whole wave-blocks: grid must be multiple of nothing special, but
+  // round to at least 8 per XCD
+  return (int)(g > 0 ? g : 1);
 }
-
-// eval fwd: same apply kernel with running stats pre-converted to mean/rstd.
-
-// ---------------------------------------------------------------------------
-// Backward stage 1: dgamma/dbeta partial reductions (ReLU-gated dy)
-// ---------------------------------------------------------------------------
-
-template <typename T>
-__global__ void bwd_stats_nhwc_kernel(const T* __restrict__ x, const T* __restrict__ dy,
-                                      const T* __restrict__ y,
-                                      const float* __restrict__ mean,
-                                      const float* __restrict__ rstd,
-                                      float* __restrict__ dg, float* __restrict__ db,
-                                      long M, int C) {
-  const long rows_per_blk = (M + gridDim.x - 1) / gridDim.x;
-  const long r0 = blockIdx.x * rows_per_blk;
-  const long r1 = min(M, r0 + rows_per_blk);
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    float sg = 0.f, sb = 0.f;
-    const float mu = mean[c], rs = rstd[c];
-    for (long r = r0; r < r1; ++r) {
-      long idx = r * C + c;
-      float g = ((float)y[idx] > 0.f) ? (float)dy[idx] : 0.f;
-      sb += g;
-      sg += g * ((float)x[idx] - mu) * rs;
-    }
-    atomicAdd(&dg[c], sg);
-    atomicAdd(&db[c], sb);
-  }
-}
-
-template <typename T, int VEC>
-__global__ void bwd_stats_nchw_kernel(const T* __restrict__ x, const T* __restrict__ dy,
-                                      const T* __restrict__ y,
-                                      const float* __restrict__ mean,
-                                      const float* __restrict__ rstd,
-                                      float* __restrict__ dg, float* __restrict__ db,
-                                      int N, int C, long HW) {
-  __shared__ float scratch[8];
-  const int c = blockIdx.x % C;
-  const int split = blockIdx.x / C;
-  const int nsplit = gridDim.x / C;
-  const float mu = mean[c], rs = rstd[c];
-  float sg = 0.f, sb = 0.f;
-  const long chunk = (HW / VEC + nsplit - 1) / nsplit;
-  const long v0 = split * chunk, v1 = min(HW / VEC, v0 + chunk);
-  for (int n = 0; n < N; ++n) {
-    const long off = ((long)n * C + c) * HW;
-    for (long i = v0 + threadIdx.x; i < v1; i += blockDim.x) {
-      #pragma unroll
-      for (int j = 0; j < VEC; ++j) {
-        long idx = off + i * VEC + j;
-        float g = ((float)y[idx] > 0.f) ? (float)dy[idx] : 0.f;
-        sb += g;
-        sg += g * ((float)x[idx] - mu) * rs;
-      }
-    }
-    if (split == 0) {
-      for (long i = (HW / VEC) * VEC + threadIdx.x; i < HW; i += blockDim.x) {
-        long idx = off + i;
-        float g = ((float)y[idx] > 0.f) ? (float)dy[idx] : 0.f;
-        sb += g;
-        sg += g * ((float)x[idx] - mu) * rs;
-      }
-    }
-  }
-  sg = block_sum<256>(sg, scratch);
-  __syncthreads();
-  sb = block_sum<256>(sb, scratch);
-  if (threadIdx.x == 0) {
-    atomicAdd(&dg[c], sg);
-    atomicAdd(&db[c], sb);
-  }
-}
-
-// Backward stage 2: dx elementwise
-template <typename T, bool NHWC>
-__global__ void bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ dy,
-                              const T* __restrict__ y,
-                              const float* __restrict__ mean,
-                              const float* __restrict__ rstd,
-                              const float* __restrict__ w,
-                              const float* __restrict__ dg,
-                              const float* __restrict__ db,
-                              T* __restrict__ dx, long total, int C, long HW,
-                              float invM) {
-  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    int c = NHWC ? (int)(idx % C) : (int)((idx / HW) % C);
-    float g = ((float)y[idx] > 0.f) ? (float)dy[idx] : 0.f;
-    float xhat = ((float)x[idx] - mean[c]) * rstd[c];
-    float v = w[c] * rstd[c] * (g - db[c] * invM - xhat * dg[c] * invM);
-    dx[idx] = (T)v;
-  }
-}
-
-// ---------------------------------------------------------------------------
-// extern "C" launchers
-// ---------------------------------------------------------------------------
 
 extern "C" {
 
 void tfosr_bn_stats(const void* x, int is_bf16, int is_nhwc, float* wsum,
                     float* wsq, int N, int C, long HW, hipStream_t s) {
   const long M = (long)N * HW;
+  const long total = M * C;
+  int vec = is_bf16 ? 8 : 4;
+  if (fast_ok(is_nhwc, total, C, vec)) {
+    u32 nvec = (u32)(total / vec);
+    int grid = fast_grid(nvec);
+    if (is_bf16)
+      hipLaunchKernelGGL(stats_nhwc_fast<bf16_t>, dim3(grid), dim3(256), 0, s,
+                         (const bf16_t*)x, wsum, wsq, nvec, (u32)C);
+    else
+      hipLaunchKernelGGL(stats_nhwc_fast<float>, dim3(grid), dim3(256), 0, s,
+                         (const float*)x, wsum, wsq, nvec, (u32)C);
+    return;
+  }
+  int grid = tfosr_grid(total, 256);
   if (is_nhwc) {
-    int grid = (int)min((long)TFOSR_MAX_GRID, (M + 63) / 64);
-    if (grid < 1) grid = 1;
     if (is_bf16)
-      hipLaunchKernelGGL(stats_nhwc_kernel<bf16_t>, dim3(grid), dim3(256), 0, s,
-                         (const bf16_t*)x, wsum, wsq, M, C);
+      hipLaunchKernelGGL((stats_generic<bf16_t, true>), dim3(grid), dim3(256), 0, s,
+                         (const bf16_t*)x, wsum, wsq, total, C, HW);
     else
-      hipLaunchKernelGGL(stats_nhwc_kernel<float>, dim3(grid), dim3(256), 0, s,
-                         (const float*)x, wsum, wsq, M, C);
+      hipLaunchKernelGGL((stats_generic<float, true>), dim3(grid), dim3(256), 0, s,
+                         (const float*)x, wsum, wsq, total, C, HW);
   } else {
-    int nsplit = max(1, min((int)(HW / (256 * 8) + 1), TFOSR_MAX_GRID / C));
-    int grid = C * nsplit;
     if (is_bf16)
-      hipLaunchKernelGGL((stats_nchw_kernel<bf16_t, 8>), dim3(grid), dim3(256), 0, s,
-                         (const bf16_t*)x, wsum, wsq, N, C, HW);
+      hipLaunchKernelGGL((stats_generic<bf16_t, false>), dim3(grid), dim3(256), 0, s,
+                         (const bf16_t*)x, wsum, wsq, total, C, HW);
     else
-      hipLaunchKernelGGL((stats_nchw_kernel<float, 4>), dim3(grid), dim3(256), 0, s,
-                         (const float*)x, wsum, wsq, N, C, HW);
+      hipLaunchKernelGGL((stats_generic<float, false>), dim3(grid), dim3(256), 0, s,
+                         (const float*)x, wsum, wsq, total, C, HW);
   }
 }
 
@@ -297,87 +357,125 @@ void tfosr_bn_finalize(const float* wsum, const float* wsq, float* save_mean,
                      M, C, momentum, eps);
 }
 
-void tfosr_bn_relu_apply(const void* x, void* y, const float* mean,
-                         const float* rstd, const float* w, const float* b,
-                         int is_bf16, int is_nhwc, long total, int C, long HW,
-                         hipStream_t s) {
-  int grid = tfosr_grid(total / 8, 256);
-  if (is_bf16) {
-    if (is_nhwc)
-      hipLaunchKernelGGL((bn_relu_apply_kernel<bf16_t, true>), dim3(grid), dim3(256),
-                         0, s, (const bf16_t*)x, (bf16_t*)y, mean, rstd, w, b,
-                         total, C, HW);
-    else
-      hipLaunchKernelGGL((bn_relu_apply_kernel<bf16_t, false>), dim3(grid), dim3(256),
-                         0, s, (const bf16_t*)x, (bf16_t*)y, mean, rstd, w, b,
-                         total, C, HW);
-  } else {
-    if (is_nhwc)
-      hipLaunchKernelGGL((bn_relu_apply_f32_kernel<true>), dim3(grid), dim3(256),
-                         0, s, (const float*)x, (float*)y, mean, rstd, w, b,
-                         total, C, HW);
-    else
-      hipLaunchKernelGGL((bn_relu_apply_f32_kernel<false>), dim3(grid), dim3(256),
-                         0, s, (const float*)x, (float*)y, mean, rstd, w, b,
-                         total, C, HW);
+// relu: 0/1; res: nullptr for plain BN
+void tfosr_bn_apply(const void* x, const void* res, void* y, const float* mean,
+                    const float* rstd, const float* w, const float* b,
+                    int is_bf16, int is_nhwc, int relu, long total, int C,
+                    long HW, hipStream_t s) {
+  int vec = is_bf16 ? 8 : 4;
+  const bool add = res != nullptr;
+  if (fast_ok(is_nhwc, total, C, vec)) {
+    u32 nvec = (u32)(total / vec);
+    int grid = fast_grid(nvec);
+#define APPLY_FAST(T, R, A) \
+    hipLaunchKernelGGL((bn_apply_fast<T, R, A>), dim3(grid), dim3(256), 0, s, \
+                       (const T*)x, (const T*)res, (T*)y, mean, rstd, w, b, \
+                       nvec, (u32)C)
+    if (is_bf16) {
+      if (relu) { if (add) APPLY_FAST(bf16_t, true, true); else APPLY_FAST(bf16_t, true, false); }
+      else      { if (add) APPLY_FAST(bf16_t, false, true); else APPLY_FAST(bf16_t, false, false); }
+    } else {
+      if (relu) { if (add) APPLY_FAST(float, true, true); else APPLY_FAST(float, true, false); }
+      else      { if (add) APPLY_FAST(float, false, true); else APPLY_FAST(float, false, false); }
+    }
+#undef APPLY_FAST
+    return;
   }
+  int grid = tfosr_grid(total, 256);
+#define APPLY_GEN(T, L, R, A) \
+  hipLaunchKernelGGL((bn_apply_generic<T, L, R, A>), dim3(grid), dim3(256), 0, s, \
+                     (const T*)x, (const T*)res, (T*)y, mean, rstd, w, b, \
+                     total, C, HW)
+  if (is_bf16) {
+    if (is_nhwc) { if (relu) { if (add) APPLY_GEN(bf16_t, true, true, true); else APPLY_GEN(bf16_t, true, true, false); }
+                   else { if (add) APPLY_GEN(bf16_t, true, false, true); else APPLY_GEN(bf16_t, true, false, false); } }
+    else { if (relu) { if (add) APPLY_GEN(bf16_t, false, true, true); else APPLY_GEN(bf16_t, false, true, false); }
+           else { if (add) APPLY_GEN(bf16_t, false, false, true); else APPLY_GEN(bf16_t, false, false, false); } }
+  } else {
+    if (is_nhwc) { if (relu) { if (add) APPLY_GEN(float, true, true, true); else APPLY_GEN(float, true, true, false); }
+                   else { if (add) APPLY_GEN(float, true, false, true); else APPLY_GEN(float, true, false, false); } }
+    else { if (relu) { if (add) APPLY_GEN(float, false, true, true); else APPLY_GEN(float, false, true, false); }
+           else { if (add) APPLY_GEN(float, false, false, true); else APPLY_GEN(float, false, false, false); } }
+  }
+#undef APPLY_GEN
 }
 
-void tfosr_bn_bwd_stats(const void* x, const void* dy, const void* y,
+// gout: non-null => write gated upstream grad (residual-branch gradient)
+void tfosr_bn_bwd_stats(const void* x, const void* dy, const void* y, void* gout,
                         const float* mean, const float* rstd, float* dg, float* db,
-                        int is_bf16, int is_nhwc, int N, int C, long HW,
+                        int is_bf16, int is_nhwc, int relu, int N, int C, long HW,
                         hipStream_t s) {
-  const long M = (long)N * HW;
-  if (is_nhwc) {
-    int grid = (int)min((long)TFOSR_MAX_GRID, (M + 63) / 64);
-    if (grid < 1) grid = 1;
-    if (is_bf16)
-      hipLaunchKernelGGL(bwd_stats_nhwc_kernel<bf16_t>, dim3(grid), dim3(256), 0, s,
-                         (const bf16_t*)x, (const bf16_t*)dy, (const bf16_t*)y,
-                         mean, rstd, dg, db, M, C);
-    else
-      hipLaunchKernelGGL(bwd_stats_nhwc_kernel<float>, dim3(grid), dim3(256), 0, s,
-                         (const float*)x, (const float*)dy, (const float*)y,
-                         mean, rstd, dg, db, M, C);
-  } else {
-    int nsplit = max(1, min((int)(HW / (256 * 8) + 1), TFOSR_MAX_GRID / C));
-    int grid = C * nsplit;
-    if (is_bf16)
-      hipLaunchKernelGGL((bwd_stats_nchw_kernel<bf16_t, 8>), dim3(grid), dim3(256),
-                         0, s, (const bf16_t*)x, (const bf16_t*)dy, (const bf16_t*)y,
-                         mean, rstd, dg, db, N, C, HW);
-    else
-      hipLaunchKernelGGL((bwd_stats_nchw_kernel<float, 4>), dim3(grid), dim3(256),
-                         0, s, (const float*)x, (const float*)dy, (const float*)y,
-                         mean, rstd, dg, db, N, C, HW);
+  const long total = (long)N * HW * C;
+  int vec = is_bf16 ? 8 : 4;
+  const bool wg = gout != nullptr;
+  if (fast_ok(is_nhwc, total, C, vec)) {
+    u32 nvec = (u32)(total / vec);
+    int grid = fast_grid(nvec);
+#define BS_FAST(T, R, W) \
+    hipLaunchKernelGGL((bwd_stats_fast<T, R, W>), dim3(grid), dim3(256), 0, s, \
+                       (const T*)x, (const T*)dy, (const T*)y, (T*)gout, \
+                       mean, rstd, dg, db, nvec, (u32)C)
+    if (is_bf16) {
+      if (relu) { if (wg) BS_FAST(bf16_t, true, true); else BS_FAST(bf16_t, true, false); }
+      else      { if (wg) BS_FAST(bf16_t, false, true); else BS_FAST(bf16_t, false, false); }
+    } else {
+      if (relu) { if (wg) BS_FAST(float, true, true); else BS_FAST(float, true, false); }
+      else      { if (wg) BS_FAST(float, false, true); else BS_FAST(float, false, false); }
+    }
+#undef BS_FAST
+    return;
   }
+  int grid = tfosr_grid(total, 256);
+#define BS_GEN(T, L, R, W) \
+  hipLaunchKernelGGL((bwd_stats_generic<T, L, R, W>), dim3(grid), dim3(256), 0, s, \
+                     (const T*)x, (const T*)dy, (const T*)y, (T*)gout, \
+                     mean, rstd, dg, db, total, C, HW)
+  if (is_bf16) {
+    if (is_nhwc) { if (relu) { if (wg) BS_GEN(bf16_t, true, true, true); else BS_GEN(bf16_t, true, true, false); }
+                   else { if (wg) BS_GEN(bf16_t, true, false, true); else BS_GEN(bf16_t, true, false, false); } }
+    else { if (relu) { if (wg) BS_GEN(bf16_t, false, true, true); else BS_GEN(bf16_t, false, true, false); }
+           else { if (wg) BS_GEN(bf16_t, false, false, true); else BS_GEN(bf16_t, false, false, false); } }
+  } else {
+    if (is_nhwc) { if (relu) { if (wg) BS_GEN(float, true, true, true); else BS_GEN(float, true, true, false); }
+                   else { if (wg) BS_GEN(float, true, false, true); else BS_GEN(float, true, false, false); } }
+    else { if (relu) { if (wg) BS_GEN(float, false, true, true); else BS_GEN(float, false, true, false); }
+           else { if (wg) BS_GEN(float, false, false, true); else BS_GEN(float, false, false, false); } }
+  }
+#undef BS_GEN
 }
 
 void tfosr_bn_bwd_dx(const void* x, const void* dy, const void* y,
                      const float* mean, const float* rstd, const float* w,
                      const float* dg, const float* db, void* dx, int is_bf16,
-                     int is_nhwc, long total, int C, long HW, hipStream_t s) {
+                     int is_nhwc, int relu, long total, int C, long HW,
+                     hipStream_t s) {
   const float invM = 1.f / (float)(total / C);
-  int grid = tfosr_grid(total, 256);
-  if (is_bf16) {
-    if (is_nhwc)
-      hipLaunchKernelGGL((bwd_dx_kernel<bf16_t, true>), dim3(grid), dim3(256), 0, s,
-                         (const bf16_t*)x, (const bf16_t*)dy, (const bf16_t*)y,
-                         mean, rstd, w, dg, db, (bf16_t*)dx, total, C, HW, invM);
-    else
-      hipLaunchKernelGGL((bwd_dx_kernel<bf16_t, false>), dim3(grid), dim3(256), 0, s,
-                         (const bf16_t*)x, (const bf16_t*)dy, (const bf16_t*)y,
-                         mean, rstd, w, dg, db, (bf16_t*)dx, total, C, HW, invM);
-  } else {
-    if (is_nhwc)
-      hipLaunchKernelGGL((bwd_dx_kernel<float, true>), dim3(grid), dim3(256), 0, s,
-                         (const float*)x, (const float*)dy, (const float*)y,
-                         mean, rstd, w, dg, db, (float*)dx, total, C, HW, invM);
-    else
-      hipLaunchKernelGGL((bwd_dx_kernel<float, false>), dim3(grid), dim3(256), 0, s,
-                         (const float*)x, (const float*)dy, (const float*)y,
-                         mean, rstd, w, dg, db, (float*)dx, total, C, HW, invM);
+  int vec = is_bf16 ? 8 : 4;
+  if (fast_ok(is_nhwc, total, C, vec)) {
+    u32 nvec = (u32)(total / vec);
+    int grid = fast_grid(nvec);
+#define DX_FAST(T, R) \
+    hipLaunchKernelGGL((bwd_dx_fast<T, R>), dim3(grid), dim3(256), 0, s, \
+                       (const T*)x, (const T*)dy, (const T*)y, (T*)dx, \
+                       mean, rstd, w, dg, db, nvec, (u32)C, invM)
+    if (is_bf16) { if (relu) DX_FAST(bf16_t, true); else DX_FAST(bf16_t, false); }
+    else { if (relu) DX_FAST(float, true); else DX_FAST(float, false); }
+#undef DX_FAST
+    return;
   }
+  int grid = tfosr_grid(total, 256);
+#define DX_GEN(T, L, R) \
+  hipLaunchKernelGGL((bwd_dx_generic<T, L, R>), dim3(grid), dim3(256), 0, s, \
+                     (const T*)x, (const T*)dy, (const T*)y, (T*)dx, \
+                     mean, rstd, w, dg, db, total, C, HW, invM)
+  if (is_bf16) {
+    if (is_nhwc) { if (relu) DX_GEN(bf16_t, true, true); else DX_GEN(bf16_t, true, false); }
+    else { if (relu) DX_GEN(bf16_t, false, true); else DX_GEN(bf16_t, false, false); }
+  } else {
+    if (is_nhwc) { if (relu) DX_GEN(float, true, true); else DX_GEN(float, true, false); }
+    else { if (relu) DX_GEN(float, false, true); else DX_GEN(float, false, false); }
+  }
+#undef DX_GEN
 }
 
 }  // extern "C"

@@ -15,7 +15,7 @@ uses.
 import torch
 import torch.nn as nn
 
-from ..ops.modules import FusedBNReLU
+from ..ops.modules import FusedBN, FusedBNAddReLU, FusedBNReLU
 
 
 def conv3x3(cin, cout, stride=1):
@@ -36,19 +36,17 @@ class Bottleneck(nn.Module):
         self.conv2 = conv3x3(width, width, stride)  # v1.5: stride on the 3x3
         self.bnrelu2 = FusedBNReLU(width)
         self.conv3 = conv1x1(width, width * self.expansion)
-        self.bn3 = nn.BatchNorm2d(width * self.expansion)
-        self.relu = nn.ReLU(inplace=True)
+        # block tail bn3 + residual-add + relu as ONE fused op
+        self.bn3 = FusedBNAddReLU(width * self.expansion)
         self.downsample = downsample
 
     def forward(self, x):
         identity = x
         out = self.bnrelu1(self.conv1(x))
         out = self.bnrelu2(self.conv2(out))
-        out = self.bn3(self.conv3(out))
         if self.downsample is not None:
             identity = self.downsample(x)
-        out += identity
-        return self.relu(out)
+        return self.bn3(self.conv3(out), identity)
 
 
 class BasicBlock(nn.Module):
@@ -59,18 +57,15 @@ class BasicBlock(nn.Module):
         self.conv1 = conv3x3(cin, width, stride)
         self.bnrelu1 = FusedBNReLU(width)
         self.conv2 = conv3x3(width, width)
-        self.bn2 = nn.BatchNorm2d(width)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn2 = FusedBNAddReLU(width)
         self.downsample = downsample
 
     def forward(self, x):
         identity = x
         out = self.bnrelu1(self.conv1(x))
-        out = self.bn2(self.conv2(out))
         if self.downsample is not None:
             identity = self.downsample(x)
-        out += identity
-        return self.relu(out)
+        return self.bn2(self.conv2(out), identity)
 
 
 class ResNet(nn.Module):
@@ -100,9 +95,6 @@ class ResNet(nn.Module):
         for m in self.modules():
             if isinstance(m, nn.Conv2d):
                 nn.init.kaiming_normal_(m.weight, mode="fan_out", nonlinearity="relu")
-            elif isinstance(m, nn.BatchNorm2d):
-                nn.init.ones_(m.weight)
-                nn.init.zeros_(m.bias)
         # zero-init last BN in each block (standard ResNet recipe)
         for m in self.modules():
             if isinstance(m, Bottleneck):
@@ -115,7 +107,7 @@ class ResNet(nn.Module):
         if stride != 1 or self.inplanes != width * block.expansion:
             downsample = nn.Sequential(
                 conv1x1(self.inplanes, width * block.expansion, stride),
-                nn.BatchNorm2d(width * block.expansion))
+                FusedBN(width * block.expansion))
         layers = [block(self.inplanes, width, stride, downsample)]
         self.inplanes = width * block.expansion
         for _ in range(1, blocks):

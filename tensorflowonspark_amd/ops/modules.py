@@ -17,31 +17,47 @@ from . import get_ext
 # Fused BatchNorm + ReLU (training fwd/bwd, inference fwd)
 # ---------------------------------------------------------------------------
 
-class _FusedBNReLUFn(torch.autograd.Function):
+class _FusedBNFn(torch.autograd.Function):
+    """BN (+residual add) (+ReLU) with fused CDNA4 kernels.
+
+    relu gating in backward uses sign(y); with the residual variant the gated
+    upstream gradient doubles as the residual-branch gradient (one pass)."""
+
     @staticmethod
-    def forward(ctx, x, weight, bias, running_mean, running_var, momentum, eps):
+    def forward(ctx, x, res, weight, bias, running_mean, running_var,
+                momentum, eps, relu):
         ext = get_ext(required=True)
-        y, save_mean, save_rstd = ext.bn_relu_fwd_train(
-            x, weight, bias, running_mean, running_var, momentum, eps)
+        y, save_mean, save_rstd = ext.bn_fwd_train(
+            x, res, weight, bias, running_mean, running_var, momentum, eps, relu)
         ctx.save_for_backward(x, y, weight, save_mean, save_rstd)
+        ctx.relu = relu
+        ctx.has_res = res is not None
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = get_ext(required=True)
         x, y, weight, save_mean, save_rstd = ctx.saved_tensors
-        dx, dweight, dbias = ext.bn_relu_bwd(
-            x, dy.contiguous(), y, weight, save_mean, save_rstd)
-        return dx, dweight, dbias, None, None, None, None
+        out = ext.bn_bwd(x, dy, y, weight, save_mean, save_rstd,
+                         ctx.relu, ctx.has_res)
+        if ctx.has_res:
+            dx, dweight, dbias, dres = out
+        else:
+            (dx, dweight, dbias), dres = out, None
+        return dx, dres, dweight, dbias, None, None, None, None, None
 
 
 class FusedBNReLU(nn.Module):
-    """BatchNorm2d + ReLU in one HBM pass each way (NCHW).
+    """BatchNorm2d (+ residual add) (+ ReLU) in one HBM pass per stage.
 
-    On MI355X this op is HBM-bandwidth-bound; fusing normalize+activation into
-    the stat kernels removes two full activation-tensor round-trips per
-    direction vs separate BN and ReLU kernels.
+    These ops are HBM-bandwidth-bound on MI355X; fusing normalize, residual
+    add and activation into the stat/normalize kernels removes whole-tensor
+    round trips vs the unfused composition, and the backward emits the
+    residual gradient from the same pass that reduces dgamma/dbeta.
     """
+
+    RELU = True
+    ADD = False
 
     def __init__(self, num_features, eps=1e-5, momentum=0.1):
         super().__init__()
@@ -54,29 +70,49 @@ class FusedBNReLU(nn.Module):
         self.register_buffer("running_var", torch.ones(num_features))
         self.register_buffer("num_batches_tracked", torch.tensor(0, dtype=torch.long))
 
-    def forward(self, x):
+    def forward(self, x, res=None):
+        assert (res is not None) == self.ADD, "residual arg mismatch"
         if x.is_cuda:
             # preserve channels_last; otherwise force standard contiguity
             if not x.is_contiguous(memory_format=torch.channels_last):
                 x = x.contiguous()
+            if res is not None:
+                if x.is_contiguous(memory_format=torch.channels_last):
+                    res = res.contiguous(memory_format=torch.channels_last)
+                else:
+                    res = res.contiguous()
             if self.training:
                 self.num_batches_tracked += 1
-                return _FusedBNReLUFn.apply(
-                    x, self.weight, self.bias,
-                    self.running_mean, self.running_var, self.momentum, self.eps)
+                return _FusedBNFn.apply(
+                    x, res, self.weight, self.bias, self.running_mean,
+                    self.running_var, self.momentum, self.eps, self.RELU)
             ext = get_ext(required=True)
             if ext is not None:
-                return ext.bn_relu_fwd_eval(
-                    x, self.weight, self.bias,
-                    self.running_mean, self.running_var, self.eps)
+                return ext.bn_fwd_eval(x, res, self.weight, self.bias,
+                                       self.running_mean, self.running_var,
+                                       self.eps, self.RELU)
         # CPU / fallback reference path
         y = F.batch_norm(x, self.running_mean, self.running_var, self.weight,
                          self.bias, self.training, self.momentum, self.eps)
-        return F.relu(y, inplace=True)
+        if res is not None:
+            y = y + res
+        return F.relu(y, inplace=True) if self.RELU else y
 
     def extra_repr(self):
-        return "{}, eps={}, momentum={}".format(
-            self.num_features, self.eps, self.momentum)
+        return "{}, eps={}, momentum={}, relu={}, add={}".format(
+            self.num_features, self.eps, self.momentum, self.RELU, self.ADD)
+
+
+class FusedBN(FusedBNReLU):
+    """Plain fused BatchNorm2d (no activation)."""
+    RELU = False
+    ADD = False
+
+
+class FusedBNAddReLU(FusedBNReLU):
+    """y = relu(bn(x) + residual) — the ResNet block tail as one op."""
+    RELU = True
+    ADD = True
 
 
 # ---------------------------------------------------------------------------

@@ -45,7 +45,7 @@ def test_bn_relu_forward_train(dtype, channels_last):
     rv = torch.ones(C, device="cuda")
     rm_ref, rv_ref = rm.clone(), rv.clone()
 
-    y, mean, rstd = ext.bn_relu_fwd_train(x, w, b, rm, rv, 0.1, 1e-5)
+    y, mean, rstd = ext.bn_fwd_train(x, None, w, b, rm, rv, 0.1, 1e-5, True)
 
     xf = x.float()
     y_ref = torch.nn.functional.relu(torch.nn.functional.batch_norm(
@@ -55,6 +55,66 @@ def test_bn_relu_forward_train(dtype, channels_last):
     _close(rm, rm_ref, 1e-3, 1e-4, "running_mean")
     _close(rv, rv_ref, 1e-2, 1e-3, "running_var")
     _close(mean, xf.mean(dim=(0, 2, 3)), 1e-3, 1e-4, "save_mean")
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("channels_last", [False, True])
+@pytest.mark.parametrize("odd_c", [False, True])
+def test_bn_add_relu_forward_backward(channels_last, odd_c):
+    """FusedBNAddReLU (block tail) vs unfused reference, incl. dres.
+    odd_c exercises the generic (non-fast-path) kernels via C=24."""
+    from tensorflowonspark_amd.ops.modules import FusedBNAddReLU
+    torch.manual_seed(7)
+    C = 24 if odd_c else 64
+    x0 = torch.randn(4, C, 8, 8)
+    r0 = torch.randn(4, C, 8, 8)
+
+    ref = FusedBNAddReLU(C)
+    x_ref = x0.clone().requires_grad_(True)
+    r_ref = r0.clone().requires_grad_(True)
+    y_ref = ref(x_ref, r_ref)
+    gy = torch.randn_like(y_ref)
+    y_ref.backward(gy)
+
+    mod = FusedBNAddReLU(C).cuda()
+    mod.load_state_dict({k: v.cuda() for k, v in ref.state_dict().items()})
+    x = x0.cuda().requires_grad_(True)
+    r = r0.cuda().requires_grad_(True)
+    xin, rin = x, r
+    if channels_last:
+        xin = x.contiguous(memory_format=torch.channels_last)
+        rin = r.contiguous(memory_format=torch.channels_last)
+    y = mod(xin, rin)
+    y.backward(gy.cuda())
+
+    _close(y, y_ref, 1e-3, 1e-4, "bn_add_relu y")
+    _close(x.grad, x_ref.grad, 1e-3, 1e-4, "bn_add_relu dx")
+    _close(r.grad, r_ref.grad, 1e-3, 1e-4, "bn_add_relu dres")
+    _close(mod.weight.grad, ref.weight.grad, 1e-3, 1e-4, "dgamma")
+    _close(mod.bias.grad, ref.bias.grad, 1e-3, 1e-4, "dbeta")
+
+
+@gpu
+@requires_gpu
+def test_fused_bn_plain_eval_and_train():
+    from tensorflowonspark_amd.ops.modules import FusedBN
+    torch.manual_seed(8)
+    C = 128
+    x0 = torch.randn(2, C, 7, 7)
+    ref = FusedBN(C)
+    mod = FusedBN(C).cuda()
+    mod.load_state_dict({k: v.cuda() for k, v in ref.state_dict().items()})
+    # train
+    y_ref = ref(x0)
+    y = mod(x0.cuda().contiguous(memory_format=torch.channels_last))
+    _close(y, y_ref, 1e-3, 1e-4, "fusedbn train y")
+    # eval
+    ref.eval()
+    mod.eval()
+    y_ref = ref(x0)
+    y = mod(x0.cuda())
+    _close(y, y_ref, 1e-3, 1e-4, "fusedbn eval y")
 
 
 @gpu
