@@ -6,15 +6,17 @@
 #include <hip/hip_runtime.h>
 
 extern "C" {
-void tfosr_bn_stats(const void*, int, int, float*, float*, int, int, long, hipStream_t);
-void tfosr_bn_finalize(const float*, const float*, float*, float*, float*, float*,
+int tfosr_bn_fast_blocks(long, int, int, int);
+void tfosr_bn_stats(const void*, int, int, float*, int, int, int, long, hipStream_t);
+void tfosr_bn_finalize(const float*, int, float*, float*, float*, float*,
                        long, int, float, float, hipStream_t);
 void tfosr_bn_apply(const void*, const void*, void*, const float*, const float*,
                     const float*, const float*, int, int, int, long, int, long,
                     hipStream_t);
 void tfosr_bn_bwd_stats(const void*, const void*, const void*, void*,
-                        const float*, const float*, float*, float*, int, int,
+                        const float*, const float*, float*, int, int, int,
                         int, int, int, long, hipStream_t);
+void tfosr_bn_bwd_merge(const float*, int, float*, float*, int, hipStream_t);
 void tfosr_bn_bwd_dx(const void*, const void*, const void*, const float*,
                      const float*, const float*, const float*, const float*,
                      void*, int, int, int, long, int, long, hipStream_t);
@@ -76,16 +78,17 @@ std::vector<at::Tensor> bn_fwd_train(at::Tensor x, c10::optional<at::Tensor> res
     res_ptr = res->data_ptr();
   }
   auto opts = x.options().dtype(at::kFloat);
-  auto wsum = at::zeros({l.C}, opts);
-  auto wsq = at::zeros({l.C}, opts);
+  long M = (long)l.N * l.HW;
+  int nb = tfosr_bn_fast_blocks(M * l.C, l.C, bf, l.nhwc);
+  auto ws = nb > 0 ? at::empty({(long)nb * 2 * l.C}, opts)
+                   : at::zeros({2L * l.C}, opts);
   auto save_mean = at::empty({l.C}, opts);
   auto save_rstd = at::empty({l.C}, opts);
   auto y = at::empty_like(x);
   auto s = cur_stream();
-  long M = (long)l.N * l.HW;
-  tfosr_bn_stats(x.data_ptr(), bf, l.nhwc, wsum.data_ptr<float>(),
-                 wsq.data_ptr<float>(), l.N, l.C, l.HW, s);
-  tfosr_bn_finalize(wsum.data_ptr<float>(), wsq.data_ptr<float>(),
+  tfosr_bn_stats(x.data_ptr(), bf, l.nhwc, ws.data_ptr<float>(), nb,
+                 l.N, l.C, l.HW, s);
+  tfosr_bn_finalize(ws.data_ptr<float>(), nb,
                     save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                     rm.data_ptr<float>(), rv.data_ptr<float>(), M, l.C,
                     (float)momentum, (float)eps, s);
@@ -117,8 +120,12 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
   int bf = dtype_flag(x);
   dy = l.nhwc ? dy.contiguous(at::MemoryFormat::ChannelsLast) : dy.contiguous();
   auto opts = x.options().dtype(at::kFloat);
-  auto dg = at::zeros({l.C}, opts);
-  auto db = at::zeros({l.C}, opts);
+  long total = (long)l.N * l.HW * l.C;
+  int nb = tfosr_bn_fast_blocks(total, l.C, bf, l.nhwc);
+  auto ws = nb > 0 ? at::empty({(long)nb * 2 * l.C}, opts)
+                   : at::zeros({2L * l.C}, opts);
+  auto dg = at::empty({l.C}, opts);
+  auto db = at::empty({l.C}, opts);
   auto dx = at::empty_like(x);
   at::Tensor gout;
   void* gout_ptr = nullptr;
@@ -129,9 +136,10 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
   auto s = cur_stream();
   tfosr_bn_bwd_stats(x.data_ptr(), dy.data_ptr(), y.data_ptr(), gout_ptr,
                      save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
-                     dg.data_ptr<float>(), db.data_ptr<float>(), bf, l.nhwc,
+                     ws.data_ptr<float>(), nb, bf, l.nhwc,
                      relu, l.N, l.C, l.HW, s);
-  long total = (long)l.N * l.HW * l.C;
+  tfosr_bn_bwd_merge(ws.data_ptr<float>(), nb, dg.data_ptr<float>(),
+                     db.data_ptr<float>(), l.C, s);
   if (grad_res) {
     // gout already carries the gated gradient: skip re-gating (and the y read)
     tfosr_bn_bwd_dx(x.data_ptr(), gout.data_ptr(), y.data_ptr(),

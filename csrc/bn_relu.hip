@@ -73,11 +73,17 @@ struct VecIO<float> {
 // channel slots) and total < 2^31
 // ---------------------------------------------------------------------------
 
+// Stage A: per-block partials into workspace[block][2C] via an LDS reduce —
+// NO global atomics (a per-thread global-atomic tail serializes ~10^5 adds
+// per channel address and was 40x slower than the loads themselves).
 template <typename T>
-__global__ void stats_nhwc_fast(const T* __restrict__ x, float* __restrict__ wsum,
-                                float* __restrict__ wsq, u32 nvec, u32 C) {
+__global__ void stats_nhwc_fast(const T* __restrict__ x,
+                                float* __restrict__ partials, u32 nvec, u32 C) {
+  extern __shared__ float lds[];  // [2C]: sums then sumsqs
   constexpr int V = VecIO<T>::V;
   const u32 c0 = ((u32)threadIdx.x * V) % C;
+  for (u32 i = threadIdx.x; i < 2 * C; i += blockDim.x) lds[i] = 0.f;
+  __syncthreads();
   float s[V], q[V], v[V];
   #pragma unroll
   for (int j = 0; j < V; ++j) { s[j] = 0.f; q[j] = 0.f; }
@@ -89,8 +95,37 @@ __global__ void stats_nhwc_fast(const T* __restrict__ x, float* __restrict__ wsu
   }
   #pragma unroll
   for (int j = 0; j < V; ++j) {
-    atomicAdd(&wsum[c0 + j], s[j]);
-    atomicAdd(&wsq[c0 + j], q[j]);
+    atomicAdd(&lds[c0 + j], s[j]);
+    atomicAdd(&lds[C + c0 + j], q[j]);
+  }
+  __syncthreads();
+  float* out = partials + (size_t)blockIdx.x * 2 * C;
+  for (u32 i = threadIdx.x; i < 2 * C; i += blockDim.x) out[i] = lds[i];
+}
+
+// Stage B: sum partials over blocks + finalize mean/rstd + running stats
+__global__ void stats_merge_finalize(const float* __restrict__ partials,
+                                     int nblocks, u32 C,
+                                     float* __restrict__ save_mean,
+                                     float* __restrict__ save_rstd,
+                                     float* __restrict__ running_mean,
+                                     float* __restrict__ running_var,
+                                     long M, float momentum, float eps) {
+  u32 c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f, q = 0.f;
+  for (int b = 0; b < nblocks; ++b) {
+    s += partials[(size_t)b * 2 * C + c];
+    q += partials[(size_t)b * 2 * C + C + c];
+  }
+  float mean = s / (float)M;
+  float var = fmaxf(q / (float)M - mean * mean, 0.f);
+  save_mean[c] = mean;
+  save_rstd[c] = rsqrtf(var + eps);
+  if (running_mean != nullptr) {
+    float unbiased = (M > 1) ? var * (float)M / (float)(M - 1) : var;
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
   }
 }
 
@@ -132,10 +167,12 @@ __global__ void bwd_stats_fast(const T* __restrict__ x, const T* __restrict__ dy
                                const T* __restrict__ y, T* __restrict__ gout,
                                const float* __restrict__ mean,
                                const float* __restrict__ rstd,
-                               float* __restrict__ dg, float* __restrict__ db,
-                               u32 nvec, u32 C) {
+                               float* __restrict__ partials, u32 nvec, u32 C) {
+  extern __shared__ float lds[];  // [2C]: dgamma then dbeta
   constexpr int V = VecIO<T>::V;
   const u32 c0 = ((u32)threadIdx.x * V) % C;
+  for (u32 i = threadIdx.x; i < 2 * C; i += blockDim.x) lds[i] = 0.f;
+  __syncthreads();
   float sg[V], sb[V], xv[V], dv[V], yv[V];
   float mu[V], rs[V];
   #pragma unroll
@@ -159,9 +196,26 @@ __global__ void bwd_stats_fast(const T* __restrict__ x, const T* __restrict__ dy
   }
   #pragma unroll
   for (int j = 0; j < V; ++j) {
-    atomicAdd(&dg[c0 + j], sg[j]);
-    atomicAdd(&db[c0 + j], sb[j]);
+    atomicAdd(&lds[c0 + j], sg[j]);
+    atomicAdd(&lds[C + c0 + j], sb[j]);
   }
+  __syncthreads();
+  float* out = partials + (size_t)blockIdx.x * 2 * C;
+  for (u32 i = threadIdx.x; i < 2 * C; i += blockDim.x) out[i] = lds[i];
+}
+
+__global__ void bwd_stats_merge(const float* __restrict__ partials, int nblocks,
+                                u32 C, float* __restrict__ dg,
+                                float* __restrict__ db) {
+  u32 c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float sg = 0.f, sb = 0.f;
+  for (int b = 0; b < nblocks; ++b) {
+    sg += partials[(size_t)b * 2 * C + c];
+    sb += partials[(size_t)b * 2 * C + C + c];
+  }
+  dg[c] = sg;
+  db[c] = sb;
 }
 
 // dx = w*rstd * (g - db/M - xhat * dg/M); g recomputed from (dy, y) or read
@@ -205,19 +259,16 @@ __global__ void bwd_dx_fast(const T* __restrict__ x, const T* __restrict__ dy,
 // ---------------------------------------------------------------------------
 
 template <typename T, bool NHWC>
-__global__ void stats_generic(const T* __restrict__ x, float* __restrict__ wsum,
-                              float* __restrict__ wsq, long total, int C, long HW) {
-  __shared__ float red[64 * 2];
-  // per-channel loop with block reduction would be slow for big C; use
-  // atomic per element strided — acceptable: this path only serves odd shapes
+__global__ void stats_generic(const T* __restrict__ x, float* __restrict__ ws,
+                              long total, int C, long HW) {
+  // atomic per element — this path only serves odd shapes/layouts
   for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
        idx += (long)gridDim.x * blockDim.x) {
     int c = NHWC ? (int)(idx % C) : (int)((idx / HW) % C);
     float v = (float)x[idx];
-    atomicAdd(&wsum[c], v);
-    atomicAdd(&wsq[c], v * v);
+    atomicAdd(&ws[c], v);
+    atomicAdd(&ws[C + c], v * v);
   }
-  (void)red;
 }
 
 template <typename T, bool NHWC, bool RELU, bool ADD>
@@ -243,15 +294,15 @@ __global__ void bwd_stats_generic(const T* __restrict__ x, const T* __restrict__
                                   const T* __restrict__ y, T* __restrict__ gout,
                                   const float* __restrict__ mean,
                                   const float* __restrict__ rstd,
-                                  float* __restrict__ dg, float* __restrict__ db,
+                                  float* __restrict__ ws,
                                   long total, int C, long HW) {
   for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
        idx += (long)gridDim.x * blockDim.x) {
     int c = NHWC ? (int)(idx % C) : (int)((idx / HW) % C);
     float g = (float)dy[idx];
     if (RELU) g = ((float)y[idx] > 0.f) ? g : 0.f;
-    atomicAdd(&db[c], g);
-    atomicAdd(&dg[c], g * ((float)x[idx] - mean[c]) * rstd[c]);
+    atomicAdd(&ws[C + c], g);
+    atomicAdd(&ws[c], g * ((float)x[idx] - mean[c]) * rstd[c]);
     if (WRITE_G) gout[idx] = (T)g;
   }
 }
@@ -275,26 +326,6 @@ __global__ void bwd_dx_generic(const T* __restrict__ x, const T* __restrict__ dy
   }
 }
 
-// finalize: mean/rstd + running-stat update (C threads)
-__global__ void stats_finalize_kernel(const float* __restrict__ wsum,
-                                      const float* __restrict__ wsq,
-                                      float* __restrict__ save_mean,
-                                      float* __restrict__ save_rstd,
-                                      float* __restrict__ running_mean,
-                                      float* __restrict__ running_var,
-                                      long M, int C, float momentum, float eps) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float mean = wsum[c] / (float)M;
-  float var = fmaxf(wsq[c] / (float)M - mean * mean, 0.f);
-  save_mean[c] = mean;
-  save_rstd[c] = rsqrtf(var + eps);
-  if (running_mean != nullptr) {
-    float unbiased = (M > 1) ? var * (float)M / (float)(M - 1) : var;
-    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
-    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
-  }
-}
 
 // ---------------------------------------------------------------------------
 // extern "C" launchers — dispatch fast path when eligible
@@ -307,54 +338,58 @@ static inline bool fast_ok(int is_nhwc, long total, int C, int vec) {
 static inline int fast_grid(long nvec) {
   long g = (nvec + 255) / 256;
   if (g > TFOSR_MAX_GRID) g = TFOSR_MAX_GRID;
-  // keep whole wave-blocks: grid must be multiple of nothing special, but
-  // round to at least 8 per XCD
   return (int)(g > 0 ? g : 1);
+}
+
+static inline int red_grid(long nvec) {
+  // reduction stage A: each thread should own >=8 vectors; cap 1024 blocks
+  long g = nvec / (256 * 8);
+  if (g < 64) g = 64;
+  if (g > 1024) g = 1024;
+  return (int)g;
 }
 
 extern "C" {
 
-void tfosr_bn_stats(const void* x, int is_bf16, int is_nhwc, float* wsum,
-                    float* wsq, int N, int C, long HW, hipStream_t s) {
-  const long M = (long)N * HW;
-  const long total = M * C;
+// #blocks the two-stage fast reduction will use; 0 => generic path
+// (caller sizes the partials workspace as max(nb,1) * 2C floats and must
+// ZERO it when nb == 0)
+int tfosr_bn_fast_blocks(long total, int C, int is_bf16, int is_nhwc) {
   int vec = is_bf16 ? 8 : 4;
-  if (fast_ok(is_nhwc, total, C, vec)) {
+  if (!fast_ok(is_nhwc, total, C, vec)) return 0;
+  return red_grid(total / vec);
+}
+
+void tfosr_bn_stats(const void* x, int is_bf16, int is_nhwc, float* partials,
+                    int nb, int N, int C, long HW, hipStream_t s) {
+  const long total = (long)N * HW * C;
+  if (nb > 0) {
+    int vec = is_bf16 ? 8 : 4;
     u32 nvec = (u32)(total / vec);
-    int grid = fast_grid(nvec);
+    size_t lds = 2 * (size_t)C * sizeof(float);
     if (is_bf16)
-      hipLaunchKernelGGL(stats_nhwc_fast<bf16_t>, dim3(grid), dim3(256), 0, s,
-                         (const bf16_t*)x, wsum, wsq, nvec, (u32)C);
+      hipLaunchKernelGGL(stats_nhwc_fast<bf16_t>, dim3(nb), dim3(256), lds, s,
+                         (const bf16_t*)x, partials, nvec, (u32)C);
     else
-      hipLaunchKernelGGL(stats_nhwc_fast<float>, dim3(grid), dim3(256), 0, s,
-                         (const float*)x, wsum, wsq, nvec, (u32)C);
+      hipLaunchKernelGGL(stats_nhwc_fast<float>, dim3(nb), dim3(256), lds, s,
+                         (const float*)x, partials, nvec, (u32)C);
     return;
   }
   int grid = tfosr_grid(total, 256);
-  if (is_nhwc) {
-    if (is_bf16)
-      hipLaunchKernelGGL((stats_generic<bf16_t, true>), dim3(grid), dim3(256), 0, s,
-                         (const bf16_t*)x, wsum, wsq, total, C, HW);
-    else
-      hipLaunchKernelGGL((stats_generic<float, true>), dim3(grid), dim3(256), 0, s,
-                         (const float*)x, wsum, wsq, total, C, HW);
-  } else {
-    if (is_bf16)
-      hipLaunchKernelGGL((stats_generic<bf16_t, false>), dim3(grid), dim3(256), 0, s,
-                         (const bf16_t*)x, wsum, wsq, total, C, HW);
-    else
-      hipLaunchKernelGGL((stats_generic<float, false>), dim3(grid), dim3(256), 0, s,
-                         (const float*)x, wsum, wsq, total, C, HW);
-  }
+#define SG(T, L) hipLaunchKernelGGL((stats_generic<T, L>), dim3(grid), dim3(256), \
+                                    0, s, (const T*)x, partials, total, C, HW)
+  if (is_bf16) { if (is_nhwc) SG(bf16_t, true); else SG(bf16_t, false); }
+  else { if (is_nhwc) SG(float, true); else SG(float, false); }
+#undef SG
 }
 
-void tfosr_bn_finalize(const float* wsum, const float* wsq, float* save_mean,
+void tfosr_bn_finalize(const float* partials, int nb, float* save_mean,
                        float* save_rstd, float* running_mean, float* running_var,
                        long M, int C, float momentum, float eps, hipStream_t s) {
   int grid = (C + 255) / 256;
-  hipLaunchKernelGGL(stats_finalize_kernel, dim3(grid), dim3(256), 0, s,
-                     wsum, wsq, save_mean, save_rstd, running_mean, running_var,
-                     M, C, momentum, eps);
+  hipLaunchKernelGGL(stats_merge_finalize, dim3(grid), dim3(256), 0, s,
+                     partials, nb > 0 ? nb : 1, (u32)C, save_mean, save_rstd,
+                     running_mean, running_var, M, momentum, eps);
 }
 
 // relu: 0/1; res: nullptr for plain BN
@@ -400,21 +435,21 @@ void tfosr_bn_apply(const void* x, const void* res, void* y, const float* mean,
 #undef APPLY_GEN
 }
 
-// gout: non-null => write gated upstream grad (residual-branch gradient)
+// gout: non-null => also write gated upstream grad (residual-branch gradient)
 void tfosr_bn_bwd_stats(const void* x, const void* dy, const void* y, void* gout,
-                        const float* mean, const float* rstd, float* dg, float* db,
-                        int is_bf16, int is_nhwc, int relu, int N, int C, long HW,
-                        hipStream_t s) {
+                        const float* mean, const float* rstd, float* partials,
+                        int nb, int is_bf16, int is_nhwc, int relu, int N, int C,
+                        long HW, hipStream_t s) {
   const long total = (long)N * HW * C;
-  int vec = is_bf16 ? 8 : 4;
   const bool wg = gout != nullptr;
-  if (fast_ok(is_nhwc, total, C, vec)) {
+  if (nb > 0) {
+    int vec = is_bf16 ? 8 : 4;
     u32 nvec = (u32)(total / vec);
-    int grid = fast_grid(nvec);
+    size_t lds = 2 * (size_t)C * sizeof(float);
 #define BS_FAST(T, R, W) \
-    hipLaunchKernelGGL((bwd_stats_fast<T, R, W>), dim3(grid), dim3(256), 0, s, \
+    hipLaunchKernelGGL((bwd_stats_fast<T, R, W>), dim3(nb), dim3(256), lds, s, \
                        (const T*)x, (const T*)dy, (const T*)y, (T*)gout, \
-                       mean, rstd, dg, db, nvec, (u32)C)
+                       mean, rstd, partials, nvec, (u32)C)
     if (is_bf16) {
       if (relu) { if (wg) BS_FAST(bf16_t, true, true); else BS_FAST(bf16_t, true, false); }
       else      { if (wg) BS_FAST(bf16_t, false, true); else BS_FAST(bf16_t, false, false); }
@@ -429,7 +464,7 @@ void tfosr_bn_bwd_stats(const void* x, const void* dy, const void* y, void* gout
 #define BS_GEN(T, L, R, W) \
   hipLaunchKernelGGL((bwd_stats_generic<T, L, R, W>), dim3(grid), dim3(256), 0, s, \
                      (const T*)x, (const T*)dy, (const T*)y, (T*)gout, \
-                     mean, rstd, dg, db, total, C, HW)
+                     mean, rstd, partials, total, C, HW)
   if (is_bf16) {
     if (is_nhwc) { if (relu) { if (wg) BS_GEN(bf16_t, true, true, true); else BS_GEN(bf16_t, true, true, false); }
                    else { if (wg) BS_GEN(bf16_t, true, false, true); else BS_GEN(bf16_t, true, false, false); } }
@@ -442,6 +477,13 @@ void tfosr_bn_bwd_stats(const void* x, const void* dy, const void* y, void* gout
            else { if (wg) BS_GEN(float, false, false, true); else BS_GEN(float, false, false, false); } }
   }
 #undef BS_GEN
+}
+
+void tfosr_bn_bwd_merge(const float* partials, int nb, float* dg, float* db,
+                        int C, hipStream_t s) {
+  int grid = (C + 255) / 256;
+  hipLaunchKernelGGL(bwd_stats_merge, dim3(grid), dim3(256), 0, s,
+                     partials, nb > 0 ? nb : 1, (u32)C, dg, db);
 }
 
 void tfosr_bn_bwd_dx(const void* x, const void* dy, const void* y,
