@@ -103,7 +103,10 @@ __global__ void stats_nhwc_fast(const T* __restrict__ x,
   for (u32 i = threadIdx.x; i < 2 * C; i += blockDim.x) out[i] = lds[i];
 }
 
-// Stage B: sum partials over blocks + finalize mean/rstd + running stats
+// Stage B: sum partials over blocks + finalize mean/rstd + running stats.
+// One block per channel, 256 threads strided over the partial blocks — a
+// single-block C-thread version serialized ~1000 loads per thread (227 us;
+// 40x this kernel's data).
 __global__ void stats_merge_finalize(const float* __restrict__ partials,
                                      int nblocks, u32 C,
                                      float* __restrict__ save_mean,
@@ -111,13 +114,17 @@ __global__ void stats_merge_finalize(const float* __restrict__ partials,
                                      float* __restrict__ running_mean,
                                      float* __restrict__ running_var,
                                      long M, float momentum, float eps) {
-  u32 c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float scratch[8];
+  const u32 c = blockIdx.x;
   float s = 0.f, q = 0.f;
-  for (int b = 0; b < nblocks; ++b) {
+  for (int b = threadIdx.x; b < nblocks; b += blockDim.x) {
     s += partials[(size_t)b * 2 * C + c];
     q += partials[(size_t)b * 2 * C + C + c];
   }
+  s = block_sum<256>(s, scratch);
+  __syncthreads();
+  q = block_sum<256>(q, scratch);
+  if (threadIdx.x != 0) return;
   float mean = s / (float)M;
   float var = fmaxf(q / (float)M - mean * mean, 0.f);
   save_mean[c] = mean;
@@ -207,15 +214,20 @@ __global__ void bwd_stats_fast(const T* __restrict__ x, const T* __restrict__ dy
 __global__ void bwd_stats_merge(const float* __restrict__ partials, int nblocks,
                                 u32 C, float* __restrict__ dg,
                                 float* __restrict__ db) {
-  u32 c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float scratch[8];
+  const u32 c = blockIdx.x;
   float sg = 0.f, sb = 0.f;
-  for (int b = 0; b < nblocks; ++b) {
+  for (int b = threadIdx.x; b < nblocks; b += blockDim.x) {
     sg += partials[(size_t)b * 2 * C + c];
     sb += partials[(size_t)b * 2 * C + C + c];
   }
-  dg[c] = sg;
-  db[c] = sb;
+  sg = block_sum<256>(sg, scratch);
+  __syncthreads();
+  sb = block_sum<256>(sb, scratch);
+  if (threadIdx.x == 0) {
+    dg[c] = sg;
+    db[c] = sb;
+  }
 }
 
 // dx = w*rstd * (g - db/M - xhat * dg/M); g recomputed from (dy, y) or read
@@ -342,8 +354,9 @@ static inline int fast_grid(long nvec) {
 }
 
 static inline int red_grid(long nvec) {
-  // reduction stage A: each thread should own >=8 vectors; cap 1024 blocks
-  long g = nvec / (256 * 8);
+  // reduction stage A: >=32 vectors per thread bounds the partials traffic
+  // to ~3% of the tensor; [64, 1024] keeps the chip busy on small layers
+  long g = nvec / (256 * 32);
   if (g < 64) g = 64;
   if (g > 1024) g = 1024;
   return (int)g;
@@ -386,8 +399,7 @@ void tfosr_bn_stats(const void* x, int is_bf16, int is_nhwc, float* partials,
 void tfosr_bn_finalize(const float* partials, int nb, float* save_mean,
                        float* save_rstd, float* running_mean, float* running_var,
                        long M, int C, float momentum, float eps, hipStream_t s) {
-  int grid = (C + 255) / 256;
-  hipLaunchKernelGGL(stats_merge_finalize, dim3(grid), dim3(256), 0, s,
+  hipLaunchKernelGGL(stats_merge_finalize, dim3(C), dim3(256), 0, s,
                      partials, nb > 0 ? nb : 1, (u32)C, save_mean, save_rstd,
                      running_mean, running_var, M, momentum, eps);
 }
@@ -481,8 +493,7 @@ void tfosr_bn_bwd_stats(const void* x, const void* dy, const void* y, void* gout
 
 void tfosr_bn_bwd_merge(const float* partials, int nb, float* dg, float* db,
                         int C, hipStream_t s) {
-  int grid = (C + 255) / 256;
-  hipLaunchKernelGGL(bwd_stats_merge, dim3(grid), dim3(256), 0, s,
+  hipLaunchKernelGGL(bwd_stats_merge, dim3(C), dim3(256), 0, s,
                      partials, nb > 0 ? nb : 1, (u32)C, dg, db);
 }
 
