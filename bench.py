@@ -33,7 +33,7 @@ def parse_args():
     p.add_argument("--batch", type=int, default=0,
                    help="per-GPU batch size (0 = auto: 256 on GPU, 8 on CPU)")
     p.add_argument("--model", default="resnet50",
-                   choices=["resnet50", "resnet56_cifar", "mnist_cnn", "unet"])
+                   choices=["resnet50", "resnet56_cifar", "mnist_cnn", "unet", "deeplabv3"])
     p.add_argument("--feed", default="spark", choices=["spark", "device"],
                    help="spark = real shm-ring ingest path; device = on-GPU synthetic")
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
@@ -43,7 +43,7 @@ def parse_args():
     return p.parse_args()
 
 
-def _feeder_main(mgr_addr, authkey, shape, batch, nclasses, stop_evt):
+def _feeder_main(mgr_addr, authkey, shape, batch, nclasses, y_shape, stop_evt):
     """Feeder process: emulates the Spark feeder task — synthetic uint8 image
     blocks + int64 labels through the manager/ring (one block per batch)."""
     from tensorflowonspark_amd import TFManager
@@ -59,7 +59,7 @@ def _feeder_main(mgr_addr, authkey, shape, batch, nclasses, stop_evt):
     pool = []
     for _ in range(4):
         x = rng.integers(0, 256, size=(batch,) + shape, dtype=np.uint8)
-        y = rng.integers(0, nclasses, size=(batch,), dtype=np.int64)
+        y = rng.integers(0, nclasses, size=(batch,) + y_shape, dtype=np.int64)
         pool.append({"x": x, "y": y})
     i = 0
     while not stop_evt.is_set():
@@ -74,7 +74,7 @@ def _feeder_main(mgr_addr, authkey, shape, batch, nclasses, stop_evt):
 class SparkFeed:
     """Worker-side consumer: DataFeed -> pinned staging -> async H2D."""
 
-    def __init__(self, shape, batch, nclasses, device, slot_bytes):
+    def __init__(self, shape, batch, nclasses, device, slot_bytes, y_shape=()):
         from tensorflowonspark_amd import TFManager, TFNode
         authkey = b"benchkey"
         self.mgr = TFManager.start(authkey, ["input", "output", "error", "free"],
@@ -92,7 +92,7 @@ class SparkFeed:
         self.stop_evt = multiprocessing.Event()
         self.proc = multiprocessing.Process(
             target=_feeder_main,
-            args=(self.mgr.address, authkey, shape, batch, nclasses, self.stop_evt),
+            args=(self.mgr.address, authkey, shape, batch, nclasses, y_shape, self.stop_evt),
             daemon=True)
         self.proc.start()
         self.feed = TFNode.DataFeed(self.mgr, train_mode=True)
@@ -108,7 +108,7 @@ class SparkFeed:
             self.nbuf = 3
             self.x_pin = [torch.empty((batch,) + shape, dtype=torch.uint8,
                                       pin_memory=True) for _ in range(self.nbuf)]
-            self.y_pin = [torch.empty((batch,), dtype=torch.int64,
+            self.y_pin = [torch.empty((batch,) + y_shape, dtype=torch.int64,
                                       pin_memory=True) for _ in range(self.nbuf)]
             self.copy_stream = torch.cuda.Stream()
             self._ready = pyq.Queue(maxsize=self.nbuf)
@@ -171,6 +171,9 @@ def build_model(name, device):
     if name == "unet":
         from tensorflowonspark_amd.models.segmentation import unet_mobilenet
         return unet_mobilenet(num_classes=3).to(device), (128, 128, 3), 3
+    if name == "deeplabv3":
+        from tensorflowonspark_amd.models.segmentation import deeplabv3_resnet50
+        return deeplabv3_resnet50(num_classes=21).to(device), (512, 512, 3), 21
     raise ValueError(name)
 
 
@@ -207,15 +210,19 @@ def main():
     amp = torch.autocast(device_type=device.type, dtype=amp_dtype,
                          enabled=args.dtype == "bf16")
 
-    slot_bytes = int(np.prod((batch,) + shape)) + batch * 8 + (1 << 16)
+    seg = args.model in ("unet", "deeplabv3")
+    y_shape = tuple(shape[:2]) if seg else ()
+    slot_bytes = (int(np.prod((batch,) + shape))
+                  + int(np.prod((batch,) + y_shape)) * 8 + (1 << 16))
     feed = None
     if args.feed == "spark":
-        feed = SparkFeed(shape, batch, nclasses, device, slot_bytes)
+        feed = SparkFeed(shape, batch, nclasses, device, slot_bytes,
+                         y_shape=y_shape)
     else:
         rng = torch.Generator(device="cpu").manual_seed(rank)
         x_dev = torch.randint(0, 256, (batch,) + shape, dtype=torch.uint8,
                               generator=rng).to(device)
-        y_dev = torch.randint(0, nclasses, (batch,), dtype=torch.int64,
+        y_dev = torch.randint(0, nclasses, (batch,) + y_shape, dtype=torch.int64,
                               generator=rng).to(device)
 
     def get_batch():
@@ -232,7 +239,12 @@ def main():
             if args.channels_last and use_cuda:
                 x = x.contiguous(memory_format=torch.channels_last)
             logits = model(x)
-            loss = softmax_cross_entropy(logits, y)
+            if seg:
+                c = logits.shape[1]
+                flat = logits.permute(0, 2, 3, 1).reshape(-1, c)
+                loss = softmax_cross_entropy(flat, y.reshape(-1))
+            else:
+                loss = softmax_cross_entropy(logits, y)
         loss.backward()
         engine.finalize_backward()
         opt.step()

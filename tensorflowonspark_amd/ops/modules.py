@@ -139,7 +139,7 @@ def softmax_cross_entropy(logits, target, reduction="mean"):
     """Sparse softmax cross-entropy (reference workloads:
     ``sparse_categorical_crossentropy``, e.g. ``mnist_spark.py:22``,
     ``resnet_cifar_dist.py:210``). Fused single-pass kernel on GPU."""
-    if logits.is_cuda and get_ext(required=True) is not None:
+    if logits.is_cuda and logits.shape[1] >= 64 and get_ext(required=True) is not None:
         loss = _SoftmaxXentFn.apply(logits.contiguous(), target.contiguous())
     else:
         loss = F.cross_entropy(logits.float(), target, reduction="none")
