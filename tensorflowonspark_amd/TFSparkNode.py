@@ -151,6 +151,28 @@ class TFNodeContext:
         import torch
         return torch.device("cuda:0") if torch.cuda.is_available() else torch.device("cpu")
 
+    # -- async (parameter-server) mode ---------------------------------------
+
+    def run_parameter_server(self):
+        """Body of a ``ps``-role ``map_fun``: serve parameter shards on this
+        node's reserved port (the one advertised in cluster_spec['ps']) until
+        the cluster shuts the node down."""
+        assert self.job_name == "ps", "run_parameter_server on non-ps node"
+        from .parallel import ps as ps_mod
+        sock = self.tmp_socket
+        self.tmp_socket = None
+        server = ps_mod.ParameterServer(sock=sock)
+        logger.info("parameter server %d serving on port %d",
+                    self.task_index, server.port)
+        server.serve_forever()
+
+    def ps_client(self):
+        """Worker-side client sharding buckets across the ps roster."""
+        from .parallel import ps as ps_mod
+        addrs = self.cluster_spec.get("ps", [])
+        assert addrs, "cluster has no ps nodes (num_ps=0?)"
+        return ps_mod.PSClient(addrs)
+
     # legacy parity
     def start_cluster_server(self, num_gpus=1, rdma=False):
         from . import TFNode
@@ -394,9 +416,10 @@ def run(fn, tf_args, cluster_meta, tensorboard=False, log_dir=None, queues=None,
         ctx._port = port
 
         release_port = bool(cluster_meta.get("release_port", True))
-        if release_port and not ctx.is_chief:
-            # chief keeps its port: it doubles as the torch.distributed
-            # rendezvous port (released inside init_process_group)
+        if release_port and not ctx.is_chief and job_name != "ps":
+            # chief keeps its port (torch.distributed rendezvous, released
+            # inside init_process_group); ps keeps its port (the parameter
+            # service binds it in ctx.run_parameter_server)
             ctx.release_port()
 
         # -- launch user fn ----------------------------------------------------
