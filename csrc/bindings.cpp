@@ -33,6 +33,16 @@ void tfosr_gemm_bt(const void*, const void*, void*, int, int, int, int,
 void tfosr_mfma_probe(const short*, const short*, float*, hipStream_t);
 }
 
+namespace tfosr {
+uint32_t crc32c(const uint8_t*, size_t, uint32_t);
+struct ScanResult {
+  std::string buffer;
+  std::vector<std::pair<size_t, size_t>> records;
+};
+ScanResult scan_file(const std::string&, bool);
+void write_file(const std::string&, const std::vector<std::string>&, bool);
+}
+
 namespace {
 
 hipStream_t cur_stream() {
@@ -256,4 +266,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bt", &gemm_bt, py::arg("a"), py::arg("b"), py::arg("out_bf16") = false);
   m.def("gemm_bf16", &gemm_bf16);
   m.def("mfma_probe", &mfma_probe);
+
+  // native TFRecord codec (CPU): bulk scan/write with HW CRC32-C
+  m.def("crc32c", [](py::bytes data) {
+    std::string s = data;
+    return tfosr::crc32c((const uint8_t*)s.data(), s.size(), 0);
+  });
+  m.def("tfrecord_read_file", [](const std::string& path, bool verify) {
+    auto res = tfosr::scan_file(path, verify);
+    py::list out;
+    for (auto& [off, len] : res.records)
+      out.append(py::bytes(res.buffer.data() + off, len));
+    return out;
+  }, py::arg("path"), py::arg("verify") = false);
+  m.def("tfrecord_write_file",
+        [](const std::string& path, const std::vector<std::string>& recs,
+           bool append) { tfosr::write_file(path, recs, append); },
+        py::arg("path"), py::arg("records"), py::arg("append") = false);
 }

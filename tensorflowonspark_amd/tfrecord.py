@@ -76,8 +76,23 @@ class TFRecordWriter:
         self.close()
 
 
+def _native_ext():
+    try:
+        from .ops import get_ext
+        return get_ext(required=False)
+    except Exception:
+        return None
+
+
 def tfrecord_iterator(path, verify=False):
-    """Yield raw record bytes from a TFRecord file."""
+    """Yield raw record bytes from a TFRecord file.
+
+    Uses the C++ codec (HW CRC32-C, single mmap-style scan) when the extension
+    is built; the pure-Python path below is the reference implementation."""
+    ext = _native_ext()
+    if ext is not None and hasattr(ext, "tfrecord_read_file"):
+        yield from ext.tfrecord_read_file(path, verify)
+        return
     with open(path, "rb") as f:
         while True:
             header = f.read(12)

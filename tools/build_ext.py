@@ -20,7 +20,7 @@ OUT_SO = os.path.join(OUT_DIR, "tfosr_hip_ops.so")
 BUILD = os.path.join(REPO, "build", "hip")
 
 SOURCES = ["bn_relu.hip", "softmax_xent.hip", "elementwise.hip",
-           "gemm_mfma.hip", "bindings.cpp"]
+           "gemm_mfma.hip", "tfrecord_codec.cpp", "bindings.cpp"]
 
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -76,5 +76,41 @@ def build(verbose=True):
     return OUT_SO
 
 
+INFER_BIN = os.path.join(REPO, "tools", "bin", "tfosr_infer")
+
+
+def build_infer_cli(verbose=True):
+    """Build the standalone C++ inference CLI (libtorch, no Python)."""
+    os.makedirs(os.path.dirname(INFER_BIN), exist_ok=True)
+    incs, libs, defs = torch_flags()
+    src = os.path.join(CSRC, "tfosr_infer.cpp")
+    codec_obj = os.path.join(BUILD, "tfrecord_codec.o")
+    deps = [src, codec_obj]
+    if (os.path.exists(INFER_BIN)
+            and all(os.path.getmtime(INFER_BIN) > os.path.getmtime(d)
+                    for d in deps)):
+        print("[build_ext] up to date:", INFER_BIN, flush=True)
+        return INFER_BIN
+    import torch.utils.cpp_extension as ce
+    rpath = ce.library_paths()[0]
+    obj = os.path.join(BUILD, "tfosr_infer.o")
+    cc = (["hipcc", "--offload-arch=" + ARCH, "-O3", "-std=c++17", "-fPIC",
+           "-c", src, "-o", obj] + incs + defs
+          + ["-Wno-deprecated-declarations"])
+    # link with g++ (hipcc would treat .o inputs as HIP sources)
+    link = (["g++", obj, codec_obj, "-o", INFER_BIN] + libs
+            + ["-L/opt/rocm/lib", "-ltorch", "-ltorch_cpu", "-lc10",
+               "-ltorch_hip", "-lc10_hip", "-lamdhip64",
+               "-Wl,-rpath," + rpath, "-Wl,-rpath,/opt/rocm/lib"])
+    for cmd in (cc, link):
+        if verbose:
+            print("[build_ext]", " ".join(cmd), flush=True)
+        subprocess.run(cmd, check=True)
+    print("[build_ext] built", INFER_BIN, flush=True)
+    return INFER_BIN
+
+
 if __name__ == "__main__":
-    sys.exit(0 if build() else 1)
+    build()
+    build_infer_cli()
+    sys.exit(0)
