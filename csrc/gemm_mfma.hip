@@ -1,19 +1,24 @@
 // MFMA bf16 GEMM for gfx950:  C[M,N] = A[M,K] @ B[N,K]^T  (both row-major,
-// K-contiguous — the natural layout for torch.nn.Linear's x @ W^T).
+// K-contiguous — the layout of torch Linear (x @ W^T) and of 1x1 convolutions
+// on channels_last tensors, where A = the activation matrix [N*H*W, Cin] and
+// B = the weight [Cout, Cin]).
 //
-// Structure (CDNA4 canonical, cf. the 128^2-tile anatomy in the CDNA HIP
-// guide §5): 256 threads = 4 waves in a 2x2 grid, each wave owns a 64x64
-// output block as 4x4 fragments of v_mfma_f32_16x16x32_bf16 accumulating in
-// AGPRs. K steps of 32: A/B tiles staged to LDS with
+// Structure (CDNA4 canonical, cf. the 128^2-tile anatomy of the CDNA HIP
+// guide §5): 256 threads = 4 waves in a WGM x WGN grid, each wave owning a
+// 64x64 output block as 4x4 fragments of v_mfma_f32_16x16x32_bf16
+// accumulating in AGPRs. K steps of 32: operand tiles staged to LDS with
 // __builtin_amdgcn_global_load_lds (16 B/lane, no VGPR round trip),
 // double-buffered; fragment reads are ds_read_b128 (8 bf16/lane).
 //
-// LDS addresses are XOR-swizzled (byte ^= ((byte>>7)&7)<<4, an involution that
-// preserves 16B chunks) so the 16-lane column reads spread across banks
-// (~2-way aliasing, which is free on CDNA4) instead of 8-way conflicting.
-// Since global_load_lds writes linearly (wave-uniform base + lane*16), the
-// swizzle is applied by pre-permuting each lane's *global source* address and
-// reading LDS through the same XOR — both-sides-or-neither.
+// Tile shapes: 128x128 (square GEMMs) and 256x64 (skinny-N: conv1x1 layers
+// with Cout == 64 would waste half of a 128-wide tile).
+//
+// LDS addresses are XOR-swizzled (byte ^= ((byte>>7)&7)<<4, an involution
+// preserving 16B chunks) so 16-lane column reads spread across banks (~2-way
+// aliasing, free on CDNA4) instead of 8-way conflicting. global_load_lds
+// writes linearly (wave-uniform base + lane*16), so the swizzle is applied by
+// pre-permuting each lane's *global source* address and reading LDS through
+// the same XOR — both-sides-or-neither.
 //
 // blockIdx -> tile mapping is XCD-aware (8 XCDs with private L2 on MI355X):
 // contiguous grid chunks land on one XCD so neighboring tiles share L2.
@@ -22,10 +27,7 @@
 typedef short bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
-#define BM 128
-#define BN 128
 #define BK 32
-#define TILE_BYTES (BM * BK * 2)  // 8192 B per operand tile
 #define LDS_SWZ(l) ((l) ^ ((((l) >> 7) & 7) << 4))
 
 __device__ __forceinline__ void stage16(const char* src,
@@ -35,11 +37,17 @@ __device__ __forceinline__ void stage16(const char* src,
       (__attribute__((address_space(3))) void*)dst, 16, 0, 0);
 }
 
-template <typename OT>
+template <typename OT, int WGM, int WGN>
 __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     OT* __restrict__ C, int M, int N, int K) {
-  __shared__ char lds[2 * 2 * TILE_BYTES];  // [buf][A|B][tile]
+  constexpr int BM = WGM * 64;
+  constexpr int BN = WGN * 64;
+  constexpr int ABYTES = BM * BK * 2;  // 64 B per row
+  constexpr int BBYTES = BN * BK * 2;
+  constexpr int ACHUNK = ABYTES / 16 / 256;  // 16B chunks per thread
+  constexpr int BCHUNK = BBYTES / 16 / 256;
+  __shared__ char lds[2 * (ABYTES + BBYTES)];
   __attribute__((address_space(3))) char* lds3 =
       (__attribute__((address_space(3))) char*)lds;
 
@@ -58,35 +66,37 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
-  const int wr = wave >> 1, wc = wave & 1;  // 2x2 wave grid
+  const int wr = wave / WGN, wc = wave % WGN;
 
-  // staging: thread t covers 16B chunks t and t+256 of each 8192B tile
-  // dest is linear; source address gets the inverse (== same) swizzle
-  long srcA_row[2], srcA_col[2], srcB_row[2], srcB_col[2];
+  // staging source addresses: dest is linear (chunk per thread), source gets
+  // the (involutive) swizzle so swizzled-read sees the right bytes
+  long srcA[ACHUNK], srcB[BCHUNK];
+  const long Kb = (long)K * 2;
   #pragma unroll
-  for (int u = 0; u < 2; ++u) {
-    int d = (t + u * 256) * 16;      // linear dest byte offset in tile
-    int sl = LDS_SWZ(d);             // source linear position
-    long row = sl >> 6;              // 64 B per row (32 bf16)
-    long col = sl & 63;              // byte within row
-    srcA_row[u] = (tile_m + row < M) ? (tile_m + row) : (M - 1);
-    srcA_col[u] = col;
-    srcB_row[u] = (tile_n + row < N) ? (tile_n + row) : (N - 1);
-    srcB_col[u] = col;
+  for (int u = 0; u < ACHUNK; ++u) {
+    int d = (t + u * 256) * 16;
+    int sl = LDS_SWZ(d);
+    long row = sl >> 6;
+    srcA[u] = ((tile_m + row < M) ? (tile_m + row) : (M - 1)) * Kb + (sl & 63);
   }
-
-  const long Kb = (long)K * 2;  // row stride in bytes
+  #pragma unroll
+  for (int u = 0; u < BCHUNK; ++u) {
+    int d = (t + u * 256) * 16;
+    int sl = LDS_SWZ(d);
+    long row = sl >> 6;
+    srcB[u] = ((tile_n + row < N) ? (tile_n + row) : (N - 1)) * Kb + (sl & 63);
+  }
 
   auto stage_tile = [&](int buf, int kt) {
     const long kbyte = (long)kt * BK * 2;
-    __attribute__((address_space(3))) char* la = lds3 + buf * 2 * TILE_BYTES;
-    __attribute__((address_space(3))) char* lb = la + TILE_BYTES;
+    __attribute__((address_space(3))) char* la = lds3 + buf * (ABYTES + BBYTES);
+    __attribute__((address_space(3))) char* lb = la + ABYTES;
     #pragma unroll
-    for (int u = 0; u < 2; ++u) {
-      int d = (t + u * 256) * 16;
-      stage16((const char*)A + srcA_row[u] * Kb + kbyte + srcA_col[u], la + d);
-      stage16((const char*)B + srcB_row[u] * Kb + kbyte + srcB_col[u], lb + d);
-    }
+    for (int u = 0; u < ACHUNK; ++u)
+      stage16((const char*)A + srcA[u] + kbyte, la + (t + u * 256) * 16);
+    #pragma unroll
+    for (int u = 0; u < BCHUNK; ++u)
+      stage16((const char*)B + srcB[u] + kbyte, lb + (t + u * 256) * 16);
   };
 
   f32x4 acc[4][4];
@@ -107,20 +117,18 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
     __syncthreads();
     if (kt + 1 < nkt) stage_tile(cur ^ 1, kt + 1);
 
-    __attribute__((address_space(3))) char* la = lds3 + cur * 2 * TILE_BYTES;
-    __attribute__((address_space(3))) char* lb = la + TILE_BYTES;
+    __attribute__((address_space(3))) char* la = lds3 + cur * (ABYTES + BBYTES);
+    __attribute__((address_space(3))) char* lb = la + ABYTES;
 
     bf16x8 afrag[4], bfrag[4];
     #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
-      int row = wr * 64 + mi * 16 + frow;
-      int l = row * 64 + kslot * 16;
+      int l = (wr * 64 + mi * 16 + frow) * 64 + kslot * 16;
       afrag[mi] = *(__attribute__((address_space(3))) bf16x8*)(la + LDS_SWZ(l));
     }
     #pragma unroll
     for (int nj = 0; nj < 4; ++nj) {
-      int row = wc * 64 + nj * 16 + frow;
-      int l = row * 64 + kslot * 16;
+      int l = (wc * 64 + nj * 16 + frow) * 64 + kslot * 16;
       bfrag[nj] = *(__attribute__((address_space(3))) bf16x8*)(lb + LDS_SWZ(l));
     }
     #pragma unroll
@@ -170,13 +178,26 @@ extern "C" {
 
 void tfosr_gemm_bt(const void* A, const void* B, void* C, int out_bf16,
                    int M, int N, int K, hipStream_t s) {
-  int ntm = (M + BM - 1) / BM, ntn = (N + BN - 1) / BN;
+  // skinny-N tile when it reduces waste (Cout=64 conv1x1 layers)
+  const bool skinny = (N % 128 != 0) && (N % 64 == 0 || N <= 64);
+  if (skinny) {
+    int ntm = (M + 255) / 256, ntn = (N + 63) / 64;
+    dim3 grid(ntm * ntn);
+    if (out_bf16)
+      hipLaunchKernelGGL((gemm_bt_kernel<bf16_t, 4, 1>), grid, dim3(256), 0, s,
+                         (const bf16_t*)A, (const bf16_t*)B, (bf16_t*)C, M, N, K);
+    else
+      hipLaunchKernelGGL((gemm_bt_kernel<float, 4, 1>), grid, dim3(256), 0, s,
+                         (const bf16_t*)A, (const bf16_t*)B, (float*)C, M, N, K);
+    return;
+  }
+  int ntm = (M + 127) / 128, ntn = (N + 127) / 128;
   dim3 grid(ntm * ntn);
   if (out_bf16)
-    hipLaunchKernelGGL(gemm_bt_kernel<bf16_t>, grid, dim3(256), 0, s,
+    hipLaunchKernelGGL((gemm_bt_kernel<bf16_t, 2, 2>), grid, dim3(256), 0, s,
                        (const bf16_t*)A, (const bf16_t*)B, (bf16_t*)C, M, N, K);
   else
-    hipLaunchKernelGGL(gemm_bt_kernel<float>, grid, dim3(256), 0, s,
+    hipLaunchKernelGGL((gemm_bt_kernel<float, 2, 2>), grid, dim3(256), 0, s,
                        (const bf16_t*)A, (const bf16_t*)B, (float*)C, M, N, K);
 }
 

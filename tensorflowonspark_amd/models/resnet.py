@@ -15,7 +15,7 @@ uses.
 import torch
 import torch.nn as nn
 
-from ..ops.modules import FusedBN, FusedBNAddReLU, FusedBNReLU
+from ..ops.modules import Conv1x1, FusedBN, FusedBNAddReLU, FusedBNReLU
 
 
 def conv3x3(cin, cout, stride=1):
@@ -23,6 +23,9 @@ def conv3x3(cin, cout, stride=1):
 
 
 def conv1x1(cin, cout, stride=1):
+    if stride == 1:
+        # pointwise convs route to the hand-written MFMA GEMM on gfx950
+        return Conv1x1(cin, cout)
     return nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
 
 

@@ -248,6 +248,85 @@ class BucketSGD:
 
 
 # ---------------------------------------------------------------------------
+# 1x1 convolution as an MFMA GEMM (channels_last)
+# ---------------------------------------------------------------------------
+
+class _Conv1x1Fn(torch.autograd.Function):
+    """Stride-1 1x1 conv on channels_last tensors == GEMM over [N*H*W, Cin].
+
+    forward / input-grad run on the hand-written gfx950 MFMA kernel
+    (C = A @ B^T, K-contiguous operands); the weight-grad contraction
+    (reduction over the huge M dim) goes through rocBLAS/hipBLASLt — a plain
+    library GEMM, per the kernel-usage policy."""
+
+    @staticmethod
+    def forward(ctx, x, weight):
+        ext = get_ext(required=True)
+        N, Cin, H, W = x.shape
+        Cout = weight.shape[0]
+        x2d = x.permute(0, 2, 3, 1).reshape(N * H * W, Cin)
+        if x2d.dtype != torch.bfloat16:
+            x2d = x2d.to(torch.bfloat16)
+        w2d = weight.view(Cout, Cin).to(torch.bfloat16)
+        y2d = ext.gemm_bt(x2d, w2d, True)
+        ctx.save_for_backward(x2d, w2d)
+        ctx.dims = (N, H, W, Cin, Cout, weight.dtype)
+        return y2d.view(N, H, W, Cout).permute(0, 3, 1, 2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_ext(required=True)
+        x2d, w2d = ctx.saved_tensors
+        N, H, W, Cin, Cout, wdtype = ctx.dims
+        dy2d = dy.permute(0, 2, 3, 1).reshape(N * H * W, Cout)
+        if not dy2d.is_contiguous():
+            dy2d = dy2d.contiguous()
+        if dy2d.dtype != torch.bfloat16:
+            dy2d = dy2d.to(torch.bfloat16)
+        # dx[M,Cin] = dy[M,Cout] @ W[Cout,Cin]  ->  gemm_bt(dy, W^T)
+        dx2d = ext.gemm_bt(dy2d, w2d.t().contiguous(), True)
+        dx = dx2d.view(N, H, W, Cin).permute(0, 3, 1, 2)
+        # dW[Cout,Cin] = dy^T @ x  (library GEMM; fp32 out for master weights)
+        dw = torch.mm(dy2d.t(), x2d).to(wdtype).view(Cout, Cin, 1, 1)
+        return dx, dw
+
+
+class Conv1x1(nn.Module):
+    """Pointwise convolution routed to the MFMA GEMM on GPU.
+
+    Backend select via TFOS_CONV1X1: 'mfma' (default), 'blas'
+    (torch.matmul / hipBLASLt for A/B comparison), 'miopen' (F.conv2d).
+    Weight kept in Conv2d's [Cout, Cin, 1, 1] shape for state_dict parity.
+    """
+
+    def __init__(self, cin, cout):
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(cout, cin, 1, 1))
+        nn.init.kaiming_normal_(self.weight, mode="fan_out", nonlinearity="relu")
+
+    def forward(self, x):
+        import os
+        backend = os.environ.get("TFOS_CONV1X1", "mfma")
+        if x.is_cuda and backend != "miopen":
+            x = x.contiguous(memory_format=torch.channels_last)
+            if backend == "blas":
+                N, Cin, H, W = x.shape
+                x2d = x.permute(0, 2, 3, 1).reshape(-1, Cin)
+                w = self.weight.view(self.weight.shape[0], Cin)
+                if x2d.dtype == torch.bfloat16:
+                    w = w.to(torch.bfloat16)
+                y2d = x2d @ w.t()
+                return y2d.view(N, H, W, -1).permute(0, 3, 1, 2)
+            if get_ext(required=True) is not None:
+                return _Conv1x1Fn.apply(x, self.weight)
+        return F.conv2d(x, self.weight.to(x.dtype))
+
+    def extra_repr(self):
+        return "{}x{} pointwise (MFMA)".format(self.weight.shape[1],
+                                               self.weight.shape[0])
+
+
+# ---------------------------------------------------------------------------
 # MFMA GEMM (bf16 inputs, fp32 accumulate) for Dense layers / serving
 # ---------------------------------------------------------------------------
 
