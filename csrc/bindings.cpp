@@ -31,6 +31,10 @@ void tfosr_sgd_step(float*, const float*, float*, float, float, float, int, long
 void tfosr_gemm_bt(const void*, const void*, void*, int, int, int, int,
                    hipStream_t);
 void tfosr_mfma_probe(const short*, const short*, float*, hipStream_t);
+void tfosr_maxpool_fwd(const void*, void*, unsigned char*, int, int, int, int,
+                       int, int, int, int, int, int, hipStream_t);
+void tfosr_maxpool_bwd(const void*, const unsigned char*, void*, int, int, int,
+                       int, int, int, int, int, int, int, hipStream_t);
 }
 
 namespace tfosr {
@@ -266,6 +270,31 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bt", &gemm_bt, py::arg("a"), py::arg("b"), py::arg("out_bf16") = false);
   m.def("gemm_bf16", &gemm_bf16);
   m.def("mfma_probe", &mfma_probe);
+  m.def("maxpool_fwd", [](at::Tensor x, long K, long S, long P) {
+    TORCH_CHECK(x.dim() == 4 && x.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "maxpool_fwd expects channels_last");
+    int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+    int OH = (H + 2 * P - K) / S + 1, OW = (W + 2 * P - K) / S + 1;
+    TORCH_CHECK(C % (x.scalar_type() == at::kBFloat16 ? 8 : 4) == 0);
+    auto y = at::empty({N, C, OH, OW}, x.options(),
+                       at::MemoryFormat::ChannelsLast);
+    auto idx = at::empty({N, OH, OW, C}, x.options().dtype(at::kByte));
+    tfosr_maxpool_fwd(x.data_ptr(), y.data_ptr(), idx.data_ptr<unsigned char>(),
+                      x.scalar_type() == at::kBFloat16, N, C, H, W, OH, OW,
+                      K, S, P, cur_stream());
+    return std::vector<at::Tensor>{y, idx};
+  });
+  m.def("maxpool_bwd", [](at::Tensor dy, at::Tensor idx, long H, long W,
+                          long K, long S, long P) {
+    dy = dy.contiguous(at::MemoryFormat::ChannelsLast);
+    int N = dy.size(0), C = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+    auto dx = at::empty({N, C, H, W}, dy.options(),
+                        at::MemoryFormat::ChannelsLast);
+    tfosr_maxpool_bwd(dy.data_ptr(), idx.data_ptr<unsigned char>(),
+                      dx.data_ptr(), dy.scalar_type() == at::kBFloat16,
+                      N, C, H, W, OH, OW, K, S, P, cur_stream());
+    return dx;
+  });
 
   // native TFRecord codec (CPU): bulk scan/write with HW CRC32-C
   m.def("crc32c", [](py::bytes data) {

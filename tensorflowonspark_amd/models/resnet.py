@@ -15,7 +15,8 @@ uses.
 import torch
 import torch.nn as nn
 
-from ..ops.modules import Conv1x1, FusedBN, FusedBNAddReLU, FusedBNReLU
+from ..ops.modules import (Conv1x1, FusedBN, FusedBNAddReLU, FusedBNReLU,
+                           FusedMaxPool2d)
 
 
 def conv3x3(cin, cout, stride=1):
@@ -85,7 +86,7 @@ class ResNet(nn.Module):
             self.stem = nn.Sequential(
                 nn.Conv2d(3, self.inplanes, 7, stride=2, padding=3, bias=False),
                 FusedBNReLU(self.inplanes),
-                nn.MaxPool2d(3, stride=2, padding=1))
+                FusedMaxPool2d(3, stride=2, padding=1))
             widths = [64, 128, 256, 512]
             strides = [1, 2, 2, 2]
         stages = []

@@ -248,6 +248,47 @@ class BucketSGD:
 
 
 # ---------------------------------------------------------------------------
+# Fused NHWC MaxPool2d (argmax saved as a packed window index)
+# ---------------------------------------------------------------------------
+
+class _MaxPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, k, s, p):
+        ext = get_ext(required=True)
+        y, idx = ext.maxpool_fwd(x, k, s, p)
+        ctx.save_for_backward(idx)
+        ctx.params = (x.shape[2], x.shape[3], k, s, p)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_ext(required=True)
+        (idx,) = ctx.saved_tensors
+        H, W, k, s, p = ctx.params
+        dx = ext.maxpool_bwd(dy, idx, H, W, k, s, p)
+        return dx, None, None, None
+
+
+class FusedMaxPool2d(nn.Module):
+    """channels_last MaxPool2d; backward is an atomic-free gather over the
+    (at most ceil(K/S)^2) windows covering each input pixel."""
+
+    def __init__(self, kernel_size=3, stride=2, padding=1):
+        super().__init__()
+        self.k, self.s, self.p = kernel_size, stride, padding
+
+    def forward(self, x):
+        if x.is_cuda and x.is_contiguous(memory_format=torch.channels_last) \
+                and x.shape[1] % (8 if x.dtype == torch.bfloat16 else 4) == 0 \
+                and get_ext(required=True) is not None:
+            return _MaxPoolFn.apply(x, self.k, self.s, self.p)
+        return F.max_pool2d(x, self.k, self.s, self.p)
+
+    def extra_repr(self):
+        return "k={}, s={}, p={}".format(self.k, self.s, self.p)
+
+
+# ---------------------------------------------------------------------------
 # 1x1 convolution as an MFMA GEMM (channels_last)
 # ---------------------------------------------------------------------------
 

@@ -293,3 +293,27 @@ def test_conv1x1_mfma(shape):
     y2 = torch.nn.functional.conv2d(x0, wref2)
     y2.backward(gy)
     _close(mod.weight.grad, wref2.grad, 3e-2, 2e-1, "conv1x1 dw")
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_maxpool(dtype):
+    from tensorflowonspark_amd.ops.modules import FusedMaxPool2d
+    torch.manual_seed(12)
+    x0 = torch.randn(3, 64, 17, 17)
+
+    x_ref = x0.clone().requires_grad_(True)
+    y_ref = torch.nn.functional.max_pool2d(x_ref, 3, 2, 1)
+    gy = torch.randn_like(y_ref)
+    y_ref.backward(gy)
+
+    mod = FusedMaxPool2d(3, 2, 1)
+    x = x0.cuda().to(dtype).contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    y = mod(x)
+    y.backward(gy.cuda().to(dtype))
+
+    tol = (1e-4, 1e-5) if dtype == torch.float32 else (2e-2, 2e-2)
+    _close(y, y_ref, *tol, "maxpool y")
+    _close(x.grad, x_ref.grad, *tol, "maxpool dx")
