@@ -31,7 +31,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--batch", type=int, default=0,
-                   help="per-GPU batch size (0 = auto: 256 on GPU, 8 on CPU)")
+                   help="per-GPU batch size (0 = auto: 1024 on GPU, 8 on CPU)")
     p.add_argument("--model", default="resnet50",
                    choices=["resnet50", "resnet56_cifar", "mnist_cnn", "unet", "deeplabv3"])
     p.add_argument("--feed", default="spark", choices=["spark", "device"],
@@ -193,7 +193,7 @@ def main():
     device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0))) \
         if use_cuda else torch.device("cpu")
 
-    batch = args.batch or (256 if use_cuda else 8)
+    batch = args.batch or (1024 if use_cuda else 8)
     model, shape, nclasses = build_model(args.model, device)
     if args.channels_last and use_cuda:
         model = model.to(memory_format=torch.channels_last)
@@ -203,8 +203,10 @@ def main():
         softmax_cross_entropy
     from tensorflowonspark_amd.parallel import DDPEngine
     engine = DDPEngine(model, bucket_mb=args.bucket_mb)
-    opt = BucketSGD(engine, lr=0.1 * world * batch / 256, momentum=0.9,
-                    weight_decay=1e-4)
+    # linear-scaling LR, capped so synthetic runs at huge global batch
+    # stay numerically sane
+    opt = BucketSGD(engine, lr=0.1 * min(world * batch, 2048) / 256,
+                    momentum=0.9, weight_decay=1e-4)
 
     amp_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
     amp = torch.autocast(device_type=device.type, dtype=amp_dtype,
