@@ -10,14 +10,17 @@ int tfosr_bn_fast_blocks(long, int, int, int);
 void tfosr_bn_stats(const void*, int, int, float*, int, int, int, long, hipStream_t);
 void tfosr_bn_finalize(const float*, int, float*, float*, float*, float*,
                        long, int, float, float, hipStream_t);
-void tfosr_bn_apply(const void*, const void*, void*, const float*, const float*,
+void tfosr_bn_apply(const void*, const void*, void*, unsigned char*,
+                    const float*, const float*,
                     const float*, const float*, int, int, int, long, int, long,
                     hipStream_t);
-void tfosr_bn_bwd_stats(const void*, const void*, const void*, void*,
+void tfosr_bn_bwd_stats(const void*, const void*, const void*,
+                        const unsigned char*, void*,
                         const float*, const float*, float*, int, int, int,
                         int, int, int, long, hipStream_t);
 void tfosr_bn_bwd_merge(const float*, int, float*, float*, int, hipStream_t);
-void tfosr_bn_bwd_dx(const void*, const void*, const void*, const float*,
+void tfosr_bn_bwd_dx(const void*, const void*, const void*,
+                     const unsigned char*, const float*,
                      const float*, const float*, const float*, const float*,
                      void*, int, int, int, long, int, long, hipStream_t);
 void tfosr_xent_fwd(const void*, const long*, float*, float*, int, int, int,
@@ -99,6 +102,16 @@ std::vector<at::Tensor> bn_fwd_train(at::Tensor x, c10::optional<at::Tensor> res
   auto save_mean = at::empty({l.C}, opts);
   auto save_rstd = at::empty({l.C}, opts);
   auto y = at::empty_like(x);
+  // relu sign bitmask (1 byte per vector) for y-free backward gating
+  at::Tensor mask;
+  unsigned char* mask_ptr = nullptr;
+  if (relu && nb > 0) {
+    long nvec = M * l.C / (bf ? 8 : 4);
+    mask = at::empty({nvec}, x.options().dtype(at::kByte));
+    mask_ptr = mask.data_ptr<unsigned char>();
+  } else {
+    mask = at::empty({0}, x.options().dtype(at::kByte));
+  }
   auto s = cur_stream();
   tfosr_bn_stats(x.data_ptr(), bf, l.nhwc, ws.data_ptr<float>(), nb,
                  l.N, l.C, l.HW, s);
@@ -106,10 +119,11 @@ std::vector<at::Tensor> bn_fwd_train(at::Tensor x, c10::optional<at::Tensor> res
                     save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                     rm.data_ptr<float>(), rv.data_ptr<float>(), M, l.C,
                     (float)momentum, (float)eps, s);
-  tfosr_bn_apply(x.data_ptr(), res_ptr, y.data_ptr(), save_mean.data_ptr<float>(),
+  tfosr_bn_apply(x.data_ptr(), res_ptr, y.data_ptr(), mask_ptr,
+                 save_mean.data_ptr<float>(),
                  save_rstd.data_ptr<float>(), w.data_ptr<float>(),
                  b.data_ptr<float>(), bf, l.nhwc, relu, M * l.C, l.C, l.HW, s);
-  return {y, save_mean, save_rstd};
+  return {y, save_mean, save_rstd, mask};
 }
 
 at::Tensor bn_fwd_eval(at::Tensor x, c10::optional<at::Tensor> res, at::Tensor w,
@@ -120,7 +134,8 @@ at::Tensor bn_fwd_eval(at::Tensor x, c10::optional<at::Tensor> res, at::Tensor w
   const void* res_ptr = res.has_value() ? res->data_ptr() : nullptr;
   auto rstd = at::rsqrt(rv + eps);
   auto y = at::empty_like(x);
-  tfosr_bn_apply(x.data_ptr(), res_ptr, y.data_ptr(), rm.data_ptr<float>(),
+  tfosr_bn_apply(x.data_ptr(), res_ptr, y.data_ptr(), nullptr,
+                 rm.data_ptr<float>(),
                  rstd.data_ptr<float>(), w.data_ptr<float>(), b.data_ptr<float>(),
                  bf, l.nhwc, relu, (long)l.N * l.HW * l.C, l.C, l.HW,
                  cur_stream());
@@ -128,7 +143,8 @@ at::Tensor bn_fwd_eval(at::Tensor x, c10::optional<at::Tensor> res, at::Tensor w
 }
 
 std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
-                               at::Tensor w, at::Tensor save_mean,
+                               at::Tensor mask, at::Tensor w,
+                               at::Tensor save_mean,
                                at::Tensor save_rstd, bool relu, bool grad_res) {
   auto l = bn_layout(x);
   int bf = dtype_flag(x);
@@ -148,7 +164,12 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
     gout_ptr = gout.data_ptr();
   }
   auto s = cur_stream();
-  tfosr_bn_bwd_stats(x.data_ptr(), dy.data_ptr(), y.data_ptr(), gout_ptr,
+  const unsigned char* mask_ptr =
+      mask.numel() ? mask.data_ptr<unsigned char>() : nullptr;
+  TORCH_CHECK(!(relu && nb > 0) || mask_ptr != nullptr,
+              "fast-path relu backward requires the forward bitmask");
+  tfosr_bn_bwd_stats(x.data_ptr(), dy.data_ptr(), y.data_ptr(), mask_ptr,
+                     gout_ptr,
                      save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                      ws.data_ptr<float>(), nb, bf, l.nhwc,
                      relu, l.N, l.C, l.HW, s);
@@ -156,13 +177,13 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
                      db.data_ptr<float>(), l.C, s);
   if (grad_res) {
     // gout already carries the gated gradient: skip re-gating (and the y read)
-    tfosr_bn_bwd_dx(x.data_ptr(), gout.data_ptr(), y.data_ptr(),
+    tfosr_bn_bwd_dx(x.data_ptr(), gout.data_ptr(), y.data_ptr(), nullptr,
                     save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                     w.data_ptr<float>(), dg.data_ptr<float>(), db.data_ptr<float>(),
                     dx.data_ptr(), bf, l.nhwc, /*relu=*/0, total, l.C, l.HW, s);
     return {dx, dg, db, gout};
   }
-  tfosr_bn_bwd_dx(x.data_ptr(), dy.data_ptr(), y.data_ptr(),
+  tfosr_bn_bwd_dx(x.data_ptr(), dy.data_ptr(), y.data_ptr(), mask_ptr,
                   save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                   w.data_ptr<float>(), dg.data_ptr<float>(), db.data_ptr<float>(),
                   dx.data_ptr(), bf, l.nhwc, relu, total, l.C, l.HW, s);

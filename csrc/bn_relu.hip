@@ -136,10 +136,11 @@ __global__ void stats_merge_finalize(const float* __restrict__ partials,
   }
 }
 
-// normalize (+add residual)(+relu); optionally store the pre-activation sum
+// normalize (+add residual)(+relu); with RELU, also emit a per-element
+// sign bitmask (1 byte per vector) so backward never re-reads y for gating
 template <typename T, bool RELU, bool ADD>
 __global__ void bn_apply_fast(const T* __restrict__ x, const T* __restrict__ res,
-                              T* __restrict__ y,
+                              T* __restrict__ y, unsigned char* __restrict__ mask,
                               const float* __restrict__ mean,
                               const float* __restrict__ rstd,
                               const float* __restrict__ w,
@@ -156,14 +157,19 @@ __global__ void bn_apply_fast(const T* __restrict__ x, const T* __restrict__ res
   for (u32 i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec; i += stride) {
     VecIO<T>::load(x + (size_t)i * V, v);
     if (ADD) VecIO<T>::load(res + (size_t)i * V, r);
+    unsigned char m = 0;
     #pragma unroll
     for (int j = 0; j < V; ++j) {
       float o = v[j] * sc[j] + sh[j];
       if (ADD) o += r[j];
-      if (RELU) o = fmaxf(o, 0.f);
+      if (RELU) {
+        if (o > 0.f) m |= (1u << j);
+        o = fmaxf(o, 0.f);
+      }
       v[j] = o;
     }
     VecIO<T>::store(y + (size_t)i * V, v);
+    if (RELU && mask != nullptr) mask[i] = m;
   }
 }
 
@@ -171,7 +177,8 @@ __global__ void bn_apply_fast(const T* __restrict__ x, const T* __restrict__ res
 // optionally write g out (the residual branch gradient for the ADD variant)
 template <typename T, bool RELU, bool WRITE_G>
 __global__ void bwd_stats_fast(const T* __restrict__ x, const T* __restrict__ dy,
-                               const T* __restrict__ y, T* __restrict__ gout,
+                               const unsigned char* __restrict__ mask,
+                               T* __restrict__ gout,
                                const float* __restrict__ mean,
                                const float* __restrict__ rstd,
                                float* __restrict__ partials, u32 nvec, u32 C) {
@@ -180,7 +187,7 @@ __global__ void bwd_stats_fast(const T* __restrict__ x, const T* __restrict__ dy
   const u32 c0 = ((u32)threadIdx.x * V) % C;
   for (u32 i = threadIdx.x; i < 2 * C; i += blockDim.x) lds[i] = 0.f;
   __syncthreads();
-  float sg[V], sb[V], xv[V], dv[V], yv[V];
+  float sg[V], sb[V], xv[V], dv[V];
   float mu[V], rs[V];
   #pragma unroll
   for (int j = 0; j < V; ++j) {
@@ -191,10 +198,10 @@ __global__ void bwd_stats_fast(const T* __restrict__ x, const T* __restrict__ dy
   for (u32 i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec; i += stride) {
     VecIO<T>::load(x + (size_t)i * V, xv);
     VecIO<T>::load(dy + (size_t)i * V, dv);
-    if (RELU) VecIO<T>::load(y + (size_t)i * V, yv);
+    unsigned char m = RELU ? mask[i] : 0;
     #pragma unroll
     for (int j = 0; j < V; ++j) {
-      float g = RELU ? (yv[j] > 0.f ? dv[j] : 0.f) : dv[j];
+      float g = RELU ? (((m >> j) & 1) ? dv[j] : 0.f) : dv[j];
       sb[j] += g;
       sg[j] += g * (xv[j] - mu[j]) * rs[j];
       dv[j] = g;
@@ -234,7 +241,8 @@ __global__ void bwd_stats_merge(const float* __restrict__ partials, int nblocks,
 // from the gated gradient written by bwd_stats (GATED=false).
 template <typename T, bool RELU>
 __global__ void bwd_dx_fast(const T* __restrict__ x, const T* __restrict__ dy,
-                            const T* __restrict__ y, T* __restrict__ dx,
+                            const unsigned char* __restrict__ mask,
+                            T* __restrict__ dx,
                             const float* __restrict__ mean,
                             const float* __restrict__ rstd,
                             const float* __restrict__ w,
@@ -243,7 +251,7 @@ __global__ void bwd_dx_fast(const T* __restrict__ x, const T* __restrict__ dy,
                             u32 nvec, u32 C, float invM) {
   constexpr int V = VecIO<T>::V;
   const u32 c0 = ((u32)threadIdx.x * V) % C;
-  float sc[V], mu[V], rs[V], dgm[V], dbm[V], xv[V], dv[V], yv[V];
+  float sc[V], mu[V], rs[V], dgm[V], dbm[V], xv[V], dv[V];
   #pragma unroll
   for (int j = 0; j < V; ++j) {
     mu[j] = mean[c0 + j]; rs[j] = rstd[c0 + j];
@@ -255,10 +263,10 @@ __global__ void bwd_dx_fast(const T* __restrict__ x, const T* __restrict__ dy,
   for (u32 i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec; i += stride) {
     VecIO<T>::load(x + (size_t)i * V, xv);
     VecIO<T>::load(dy + (size_t)i * V, dv);
-    if (RELU) VecIO<T>::load(y + (size_t)i * V, yv);
+    unsigned char m = RELU ? mask[i] : 0;
     #pragma unroll
     for (int j = 0; j < V; ++j) {
-      float g = RELU ? (yv[j] > 0.f ? dv[j] : 0.f) : dv[j];
+      float g = RELU ? (((m >> j) & 1) ? dv[j] : 0.f) : dv[j];
       float xhat = (xv[j] - mu[j]) * rs[j];
       dv[j] = sc[j] * (g - dbm[j] - xhat * dgm[j]);
     }
@@ -404,8 +412,9 @@ void tfosr_bn_finalize(const float* partials, int nb, float* save_mean,
                      running_mean, running_var, M, momentum, eps);
 }
 
-// relu: 0/1; res: nullptr for plain BN
-void tfosr_bn_apply(const void* x, const void* res, void* y, const float* mean,
+// relu: 0/1; res: nullptr for plain BN; mask: sign bits out (fast path only)
+void tfosr_bn_apply(const void* x, const void* res, void* y, unsigned char* mask,
+                    const float* mean,
                     const float* rstd, const float* w, const float* b,
                     int is_bf16, int is_nhwc, int relu, long total, int C,
                     long HW, hipStream_t s) {
@@ -416,7 +425,7 @@ void tfosr_bn_apply(const void* x, const void* res, void* y, const float* mean,
     int grid = fast_grid(nvec);
 #define APPLY_FAST(T, R, A) \
     hipLaunchKernelGGL((bn_apply_fast<T, R, A>), dim3(grid), dim3(256), 0, s, \
-                       (const T*)x, (const T*)res, (T*)y, mean, rstd, w, b, \
+                       (const T*)x, (const T*)res, (T*)y, mask, mean, rstd, w, b, \
                        nvec, (u32)C)
     if (is_bf16) {
       if (relu) { if (add) APPLY_FAST(bf16_t, true, true); else APPLY_FAST(bf16_t, true, false); }
@@ -448,7 +457,9 @@ void tfosr_bn_apply(const void* x, const void* res, void* y, const float* mean,
 }
 
 // gout: non-null => also write gated upstream grad (residual-branch gradient)
-void tfosr_bn_bwd_stats(const void* x, const void* dy, const void* y, void* gout,
+// fast path gates from `mask` (written by apply); generic path gates from y
+void tfosr_bn_bwd_stats(const void* x, const void* dy, const void* y,
+                        const unsigned char* mask, void* gout,
                         const float* mean, const float* rstd, float* partials,
                         int nb, int is_bf16, int is_nhwc, int relu, int N, int C,
                         long HW, hipStream_t s) {
@@ -460,7 +471,7 @@ void tfosr_bn_bwd_stats(const void* x, const void* dy, const void* y, void* gout
     size_t lds = 2 * (size_t)C * sizeof(float);
 #define BS_FAST(T, R, W) \
     hipLaunchKernelGGL((bwd_stats_fast<T, R, W>), dim3(nb), dim3(256), lds, s, \
-                       (const T*)x, (const T*)dy, (const T*)y, (T*)gout, \
+                       (const T*)x, (const T*)dy, mask, (T*)gout, \
                        mean, rstd, partials, nvec, (u32)C)
     if (is_bf16) {
       if (relu) { if (wg) BS_FAST(bf16_t, true, true); else BS_FAST(bf16_t, true, false); }
@@ -498,6 +509,7 @@ void tfosr_bn_bwd_merge(const float* partials, int nb, float* dg, float* db,
 }
 
 void tfosr_bn_bwd_dx(const void* x, const void* dy, const void* y,
+                     const unsigned char* mask,
                      const float* mean, const float* rstd, const float* w,
                      const float* dg, const float* db, void* dx, int is_bf16,
                      int is_nhwc, int relu, long total, int C, long HW,
@@ -509,7 +521,7 @@ void tfosr_bn_bwd_dx(const void* x, const void* dy, const void* y,
     int grid = fast_grid(nvec);
 #define DX_FAST(T, R) \
     hipLaunchKernelGGL((bwd_dx_fast<T, R>), dim3(grid), dim3(256), 0, s, \
-                       (const T*)x, (const T*)dy, (const T*)y, (T*)dx, \
+                       (const T*)x, (const T*)dy, mask, (T*)dx, \
                        mean, rstd, w, dg, db, nvec, (u32)C, invM)
     if (is_bf16) { if (relu) DX_FAST(bf16_t, true); else DX_FAST(bf16_t, false); }
     else { if (relu) DX_FAST(float, true); else DX_FAST(float, false); }

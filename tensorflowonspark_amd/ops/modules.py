@@ -27,9 +27,9 @@ class _FusedBNFn(torch.autograd.Function):
     def forward(ctx, x, res, weight, bias, running_mean, running_var,
                 momentum, eps, relu):
         ext = get_ext(required=True)
-        y, save_mean, save_rstd = ext.bn_fwd_train(
+        y, save_mean, save_rstd, mask = ext.bn_fwd_train(
             x, res, weight, bias, running_mean, running_var, momentum, eps, relu)
-        ctx.save_for_backward(x, y, weight, save_mean, save_rstd)
+        ctx.save_for_backward(x, y, mask, weight, save_mean, save_rstd)
         ctx.relu = relu
         ctx.has_res = res is not None
         return y
@@ -37,8 +37,8 @@ class _FusedBNFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         ext = get_ext(required=True)
-        x, y, weight, save_mean, save_rstd = ctx.saved_tensors
-        out = ext.bn_bwd(x, dy, y, weight, save_mean, save_rstd,
+        x, y, mask, weight, save_mean, save_rstd = ctx.saved_tensors
+        out = ext.bn_bwd(x, dy, y, mask, weight, save_mean, save_rstd,
                          ctx.relu, ctx.has_res)
         if ctx.has_res:
             dx, dweight, dbias, dres = out
