@@ -132,10 +132,17 @@ class TFCluster(object):
                 raise Exception("cluster startup/runtime error: {}".format(
                     tf_status["error"]))
 
-            # stop ps/evaluator nodes: connect to their remote managers from
-            # the driver and put None on 'control' (reference TFCluster.py:186-194)
+            # driver-thread ps nodes stop in-process
+            for srv in getattr(self, "_driver_ps", []):
+                srv.stop()
+
+            # stop executor ps/evaluator nodes: connect to their remote
+            # managers from the driver and put None on 'control'
+            # (reference TFCluster.py:186-194)
             from . import TFManager
             for node in ps_eval:
+                if node.get("driver_ps"):
+                    continue
                 try:
                     m = TFManager.connect(tuple(node["addr"]),
                                           bytes.fromhex(node["authkey"]))
@@ -221,8 +228,32 @@ def run(sc, map_fun, tf_args, num_executors, num_ps=0, tensorboard=False,
 
     tf_status.clear()
 
+    # -- driver-side ps nodes (reference TFCluster.py:298-316) ----------------
+    driver_ps = []
+    executor_range = range(num_executors)
+    if driver_ps_nodes and num_ps > 0:
+        import threading as _threading
+
+        from . import util as _util
+        from .parallel import ps as _ps_mod
+        host = _util.get_ip_address()
+        if os.environ.get("TFOS_FORCE_LOOPBACK"):
+            host = "127.0.0.1"
+        for i, ps_id in enumerate(cluster_template.get("ps", [])):
+            server_ps = _ps_mod.ParameterServer(port=0)
+            t_ps = _threading.Thread(target=server_ps.serve_forever, daemon=True)
+            t_ps.start()
+            meta = {"executor_id": ps_id, "host": host, "job_name": "ps",
+                    "task_index": i, "port": server_ps.port, "tb_pid": 0,
+                    "tb_port": None, "addr": None, "authkey": "",
+                    "driver_ps": True}
+            server.reservations.add(meta)
+            driver_ps.append(server_ps)
+        executor_range = range(num_ps, num_executors)
+        logger.info("running %d ps node(s) as driver threads", num_ps)
+
     # -- launch bootstrap job on a daemon thread ------------------------------
-    nodeRDD = sc.parallelize(range(num_executors), num_executors)
+    nodeRDD = sc.parallelize(executor_range, len(executor_range))
     background = (input_mode == InputMode.SPARK)
 
     def _start(status):
@@ -266,4 +297,5 @@ def run(sc, map_fun, tf_args, num_executors, num_ps=0, tensorboard=False,
     cluster.queues = queues
     cluster.server = server
     cluster._start_thread = t
+    cluster._driver_ps = driver_ps
     return cluster
