@@ -45,7 +45,7 @@ def test_bn_relu_forward_train(dtype, channels_last):
     rv = torch.ones(C, device="cuda")
     rm_ref, rv_ref = rm.clone(), rv.clone()
 
-    y, mean, rstd = ext.bn_fwd_train(x, None, w, b, rm, rv, 0.1, 1e-5, True)
+    y, mean, rstd, _mask = ext.bn_fwd_train(x, None, w, b, rm, rv, 0.1, 1e-5, True)
 
     xf = x.float()
     y_ref = torch.nn.functional.relu(torch.nn.functional.batch_norm(
