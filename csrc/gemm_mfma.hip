@@ -23,6 +23,8 @@
 // blockIdx -> tile mapping is XCD-aware (8 XCDs with private L2 on MI355X):
 // contiguous grid chunks land on one XCD so neighboring tiles share L2.
 #include "tfosr_common.h"
+#include <cstdlib>
+#include <cstring>
 
 typedef short bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
@@ -160,6 +162,141 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// 256x256-tile, 8-wave, 4-deep-pipelined variant.
+//
+// The 2-buffer kernel above drains vmcnt to 0 at every K-step barrier — the
+// classic CDNA4 stall (the s_waitcnt vmcnt(0) before s_barrier serializes the
+// staging queue against compute). This variant keeps a 4-deep LDS ring of
+// K-step tiles (4 x 32 KB = 128 KB) and waits with a *counted* vmcnt:
+//
+//   step t: s_waitcnt vmcnt(8); s_barrier;       // tile t's 4 loads landed
+//           stage(t+3) -> ring[(t+3)%4];          // overwrites ring slot of
+//                                                  // t-1, fully read last step
+//           12 x ds_read_b128 frags; 32 MFMA (setprio-wrapped); s_barrier;
+//
+// Race-freedom: loads issued at step t target ring[(t+3)%4] == ring[(t-1)%4],
+// whose reads completed at step t-1's closing barrier; vmcnt completion is
+// in-order per thread, so "<= 8 outstanding" proves the 4 oldest (tile t's)
+// have written LDS. No vmcnt(0) anywhere in the main loop.
+// ---------------------------------------------------------------------------
+
+#define B2_BM 256
+#define B2_BN 256
+#define B2_TILE_BYTES (B2_BM * BK * 2)   // 16 KB per operand per K-step
+#define B2_SLOT (2 * B2_TILE_BYTES)      // A+B per ring slot
+
+template <typename OT>
+__global__ __launch_bounds__(512, 1) void gemm_bt256_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    OT* __restrict__ C, int M, int N, int K) {
+  __shared__ char lds[4 * B2_SLOT];  // 128 KB ring
+  __attribute__((address_space(3))) char* lds3 =
+      (__attribute__((address_space(3))) char*)lds;
+
+  const int ntn = (N + B2_BN - 1) / B2_BN;
+  const int nwg = gridDim.x;
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const long tile_m = (long)(wgid / ntn) * B2_BM;
+  const long tile_n = (long)(wgid % ntn) * B2_BN;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;          // 8 waves: 2 (M) x 4 (N)
+  const int wr = wave >> 2;         // 0..1 -> 128 rows each
+  const int wc = wave & 3;          // 0..3 -> 64 cols each
+
+  // staging: 4 x 16B chunks per thread per K-step (2 for A, 2 for B)
+  const long Kb = (long)K * 2;
+  long srcA[2], srcB[2];
+  #pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    int d = (t + u * 512) * 16;
+    int sl = LDS_SWZ(d);
+    long row = sl >> 6;
+    srcA[u] = ((tile_m + row < M) ? (tile_m + row) : (M - 1)) * Kb + (sl & 63);
+    srcB[u] = ((tile_n + row < N) ? (tile_n + row) : (N - 1)) * Kb + (sl & 63);
+  }
+
+  auto stage_tile = [&](int slot, int kt) {
+    const long kbyte = (long)kt * BK * 2;
+    __attribute__((address_space(3))) char* la = lds3 + slot * B2_SLOT;
+    __attribute__((address_space(3))) char* lb = la + B2_TILE_BYTES;
+    #pragma unroll
+    for (int u = 0; u < 2; ++u)
+      stage16((const char*)A + srcA[u] + kbyte, la + (t + u * 512) * 16);
+    #pragma unroll
+    for (int u = 0; u < 2; ++u)
+      stage16((const char*)B + srcB[u] + kbyte, lb + (t + u * 512) * 16);
+  };
+
+  f32x4 acc[8][4];
+  #pragma unroll
+  for (int i = 0; i < 8; ++i)
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nkt = K / BK;
+  stage_tile(0, 0);
+  if (1 < nkt) stage_tile(1, 1);
+  if (2 < nkt) stage_tile(2, 2);
+
+  const int frow = lane & 15;
+  const int kslot = lane >> 4;
+
+  for (int kt = 0; kt < nkt; ++kt) {
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __syncthreads();
+    if (kt + 3 < nkt) stage_tile((kt + 3) & 3, kt + 3);
+
+    __attribute__((address_space(3))) char* la = lds3 + (kt & 3) * B2_SLOT;
+    __attribute__((address_space(3))) char* lb = la + B2_TILE_BYTES;
+
+    bf16x8 afrag[8], bfrag[4];
+    #pragma unroll
+    for (int nj = 0; nj < 4; ++nj) {
+      int l = (wc * 64 + nj * 16 + frow) * 64 + kslot * 16;
+      bfrag[nj] = *(__attribute__((address_space(3))) bf16x8*)(lb + LDS_SWZ(l));
+    }
+    #pragma unroll
+    for (int mi = 0; mi < 8; ++mi) {
+      int l = (wr * 128 + mi * 16 + frow) * 64 + kslot * 16;
+      afrag[mi] = *(__attribute__((address_space(3))) bf16x8*)(la + LDS_SWZ(l));
+    }
+    __builtin_amdgcn_s_setprio(1);
+    #pragma unroll
+    for (int mi = 0; mi < 8; ++mi)
+      #pragma unroll
+      for (int nj = 0; nj < 4; ++nj)
+        acc[mi][nj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[mi], bfrag[nj], acc[mi][nj], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+  }
+
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+  #pragma unroll
+  for (int mi = 0; mi < 8; ++mi) {
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      long m = tile_m + wr * 128 + mi * 16 + crow0 + r;
+      if (m >= M) continue;
+      #pragma unroll
+      for (int nj = 0; nj < 4; ++nj) {
+        long n = tile_n + wc * 64 + nj * 16 + ccol;
+        if (n < N) C[m * N + n] = (OT)acc[mi][nj][r];
+      }
+    }
+  }
+}
+
 // Raw-intrinsic probe: feed per-lane fragments, observe the accumulator —
 // lets the host test discover/verify the A/B/C lane->element mappings.
 __global__ void mfma_probe_kernel(const short* __restrict__ a,
@@ -178,6 +315,23 @@ extern "C" {
 
 void tfosr_gemm_bt(const void* A, const void* B, void* C, int out_bf16,
                    int M, int N, int K, hipStream_t s) {
+  // 256^2 4-deep-pipelined kernel for large tiles (toggle: TFOS_GEMM256=off)
+  static int use256 = -1;
+  if (use256 < 0) {
+    const char* e = getenv("TFOS_GEMM256");
+    use256 = (e == nullptr || strcmp(e, "off") != 0) ? 1 : 0;
+  }
+  if (use256 && M >= 1024 && N >= 192 && K >= 128) {
+    int ntm = (M + B2_BM - 1) / B2_BM, ntn = (N + B2_BN - 1) / B2_BN;
+    dim3 grid(ntm * ntn);
+    if (out_bf16)
+      hipLaunchKernelGGL(gemm_bt256_kernel<bf16_t>, grid, dim3(512), 0, s,
+                         (const bf16_t*)A, (const bf16_t*)B, (bf16_t*)C, M, N, K);
+    else
+      hipLaunchKernelGGL(gemm_bt256_kernel<float>, grid, dim3(512), 0, s,
+                         (const bf16_t*)A, (const bf16_t*)B, (float*)C, M, N, K);
+    return;
+  }
   // skinny-N tile when it reduces waste (Cout=64 conv1x1 layers)
   const bool skinny = (N % 128 != 0) && (N % 64 == 0 || N <= 64);
   if (skinny) {
