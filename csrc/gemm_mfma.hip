@@ -315,13 +315,15 @@ extern "C" {
 
 void tfosr_gemm_bt(const void* A, const void* B, void* C, int out_bf16,
                    int M, int N, int K, hipStream_t s) {
-  // 256^2 4-deep-pipelined kernel for large tiles (toggle: TFOS_GEMM256=off)
+  // 256^2 4-deep-pipelined kernel for large-K tiles (917 TF @4096^3 vs 644
+  // for the 2-buffer kernel; the deep prologue loses on the skinny-K conv
+  // shapes, which stay on the 128^2/256x64 kernels). Toggle: TFOS_GEMM256=off
   static int use256 = -1;
   if (use256 < 0) {
     const char* e = getenv("TFOS_GEMM256");
     use256 = (e == nullptr || strcmp(e, "off") != 0) ? 1 : 0;
   }
-  if (use256 && M >= 1024 && N >= 192 && K >= 128) {
+  if (use256 && M >= 1024 && N >= 192 && K >= 1024) {
     int ntm = (M + B2_BM - 1) / B2_BM, ntn = (N + B2_BN - 1) / B2_BN;
     dim3 grid(ntm * ntn);
     if (out_bf16)
