@@ -210,6 +210,70 @@ class _Job:
         return self.fn
 
 
+class LocalDStream:
+    """Minimal DStream stand-in: RDDs pushed via the owning streaming context
+    flow to every registered foreachRDD callback."""
+
+    def __init__(self, ssc):
+        self.ssc = ssc
+        self._callbacks = []
+
+    def foreachRDD(self, fn):
+        self._callbacks.append(fn)
+
+
+class LocalStreamingContext:
+    """Spark-Streaming stand-in for the DStream feed path
+    (reference streaming example: ``mnist_spark_streaming.py``).
+
+    ``queueStream()`` returns a stream; ``push(rdd)`` delivers a micro-batch
+    to the registered callbacks on a worker thread; ``stop()`` /
+    ``awaitTerminationOrTimeout`` mirror the pyspark surface TFCluster uses.
+    """
+
+    def __init__(self, sc):
+        self.sc = sc
+        self._queue = pyqueue.Queue()
+        self._streams = []
+        self._stopped = threading.Event()
+        self._thread = threading.Thread(target=self._loop, daemon=True)
+        self._thread.start()
+
+    def queueStream(self):
+        s = LocalDStream(self)
+        self._streams.append(s)
+        return s
+
+    def push(self, rdd):
+        self._queue.put(rdd)
+
+    def _loop(self):
+        while not self._stopped.is_set():
+            try:
+                rdd = self._queue.get(timeout=0.2)
+            except pyqueue.Empty:
+                continue
+            for stream in self._streams:
+                for cb in stream._callbacks:
+                    try:
+                        cb(rdd)
+                    except Exception:
+                        logger.exception("streaming callback failed")
+
+    def awaitTerminationOrTimeout(self, timeout):
+        return self._stopped.wait(timeout)
+
+    def stop(self, stopSparkContext=False, stopGraceFully=True):
+        if stopGraceFully:
+            # drain pending micro-batches first
+            while not self._queue.empty() and not self._stopped.is_set():
+                import time as _t
+                _t.sleep(0.1)
+        self._stopped.set()
+        if stopSparkContext:
+            self.sc.stop()
+
+
 class LocalSparkContext:
     """N persistent executor processes with Spark-like task dispatch."""
 
