@@ -157,3 +157,40 @@ def loadTFRecords(sc, input_dir, binary_features=None):
             "{}:{}".format(n, t) for n, t in schema))
     loadedDF[id(df)] = input_dir
     return df
+
+
+def parse_schema(simple_string):
+    """Parse a Spark-SQL simpleString like ``struct<a:bigint,b:array<double>>``
+    into a [(name, dtype)] schema (parity: reference SimpleTypeParser.scala:28
+    — base types binary/boolean/int/long/bigint/float/double/string plus 1-D
+    arrays)."""
+    s = simple_string.strip()
+    if s.startswith("struct<") and s.endswith(">"):
+        s = s[len("struct<"):-1]
+    fields = []
+    depth = 0
+    token = ""
+    parts = []
+    for ch in s:
+        if ch == "<":
+            depth += 1
+        elif ch == ">":
+            depth -= 1
+        if ch == "," and depth == 0:
+            parts.append(token)
+            token = ""
+        else:
+            token += ch
+    if token.strip():
+        parts.append(token)
+    valid = {"binary", "boolean", "int", "long", "bigint", "float", "double",
+             "string", "tinyint", "smallint"}
+    for part in parts:
+        name, _, dtype = part.partition(":")
+        name, dtype = name.strip(), dtype.strip()
+        base = dtype[6:-1] if dtype.startswith("array<") and dtype.endswith(">") \
+            else dtype
+        if base not in valid:
+            raise ValueError("unsupported type '{}' in schema".format(dtype))
+        fields.append((name, dtype))
+    return fields
