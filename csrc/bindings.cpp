@@ -38,6 +38,8 @@ void tfosr_maxpool_fwd(const void*, void*, unsigned char*, int, int, int, int,
                        int, int, int, int, int, int, hipStream_t);
 void tfosr_maxpool_bwd(const void*, const unsigned char*, void*, int, int, int,
                        int, int, int, int, int, int, int, hipStream_t);
+void tfosr_conv3x3(const void*, const void*, const void*, void*, int, int, int,
+                   int, int, int, int, int, int, int, hipStream_t);
 }
 
 namespace tfosr {
@@ -304,6 +306,24 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                       x.scalar_type() == at::kBFloat16, N, C, H, W, OH, OW,
                       K, S, P, cur_stream());
     return std::vector<at::Tensor>{y, idx};
+  });
+  m.def("conv3x3_fwd", [](at::Tensor x, at::Tensor w9, long Cout, long S,
+                          long P) {
+    TORCH_CHECK(x.dim() == 4 && x.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "conv3x3_fwd expects channels_last input");
+    TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+                w9.scalar_type() == at::kBFloat16);
+    int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+    TORCH_CHECK(Cin % 32 == 0, "conv3x3 requires Cin % 32 == 0");
+    TORCH_CHECK(w9.size(1) == 9 * Cin);
+    int OH = (H + 2 * P - 3) / S + 1, OW = (W + 2 * P - 3) / S + 1;
+    auto y = at::empty({N, Cout, OH, OW}, x.options(),
+                       at::MemoryFormat::ChannelsLast);
+    auto guard = at::zeros({64}, x.options());
+    tfosr_conv3x3(x.data_ptr(), w9.contiguous().data_ptr(), guard.data_ptr(),
+                  y.data_ptr(), /*out_bf16=*/1, N, H, W, Cin, Cout, OH, OW,
+                  S, P, cur_stream());
+    return y;
   });
   m.def("maxpool_bwd", [](at::Tensor dy, at::Tensor idx, long H, long W,
                           long K, long S, long P) {

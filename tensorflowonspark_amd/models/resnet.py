@@ -15,11 +15,14 @@ uses.
 import torch
 import torch.nn as nn
 
-from ..ops.modules import (Conv1x1, FusedBN, FusedBNAddReLU, FusedBNReLU,
-                           FusedMaxPool2d)
+from ..ops.modules import (Conv1x1, Conv3x3, FusedBN, FusedBNAddReLU,
+                           FusedBNReLU, FusedMaxPool2d)
 
 
 def conv3x3(cin, cout, stride=1):
+    if stride == 1 and cin % 32 == 0 and cout >= 64:
+        # stride-1 3x3s route to the implicit-GEMM MFMA kernel on gfx950
+        return Conv3x3(cin, cout)
     return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
 
 

@@ -374,6 +374,71 @@ class Conv1x1(nn.Module):
 
 
 # ---------------------------------------------------------------------------
+# 3x3 convolution as an implicit MFMA GEMM (channels_last, stride 1)
+# ---------------------------------------------------------------------------
+
+class _Conv3x3Fn(torch.autograd.Function):
+    """Stride-1 3x3 conv on channels_last bf16 via the 4-deep-pipelined
+    implicit-GEMM kernel. dgrad = conv3x3 of dy with the flipped/transposed
+    weight (same kernel); wrw goes through MIOpen's tuned igemm."""
+
+    @staticmethod
+    def forward(ctx, x, weight):
+        ext = get_ext(required=True)
+        Cout, Cin = weight.shape[0], weight.shape[1]
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        w9 = weight.permute(0, 2, 3, 1).reshape(Cout, 9 * Cin)             .to(torch.bfloat16).contiguous()
+        y = ext.conv3x3_fwd(x, w9, Cout, 1, 1)
+        ctx.save_for_backward(x, weight)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_ext(required=True)
+        x, weight = ctx.saved_tensors
+        Cout, Cin = weight.shape[0], weight.shape[1]
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        if dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
+        # W'[cin][r][s][cout] = W[cout][2-r][2-s][cin]
+        w9p = weight.flip(2, 3).permute(1, 2, 3, 0)             .reshape(Cin, 9 * Cout).to(torch.bfloat16).contiguous()
+        dx = ext.conv3x3_fwd(dy, w9p, Cin, 1, 1)
+        w4 = weight.to(torch.bfloat16).contiguous(
+            memory_format=torch.channels_last)
+        _, dw, _ = torch.ops.aten.convolution_backward(
+            dy, x, w4, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+            [False, True, False])
+        return dx, dw.to(weight.dtype)
+
+
+class Conv3x3(nn.Module):
+    """3x3/stride-1/pad-1 conv routed to the implicit-GEMM MFMA kernel.
+
+    Backend select via TFOS_CONV3X3: 'mfma' (default), 'miopen'.
+    Weight in Conv2d's [Cout, Cin, 3, 3] shape for state_dict parity.
+    """
+
+    def __init__(self, cin, cout):
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(cout, cin, 3, 3))
+        nn.init.kaiming_normal_(self.weight, mode="fan_out", nonlinearity="relu")
+        self._eligible = cin % 32 == 0 and cout % 32 == 0 and cout >= 64
+
+    def forward(self, x):
+        import os
+        backend = os.environ.get("TFOS_CONV3X3", "mfma")
+        if x.is_cuda and self._eligible and backend == "mfma"                 and get_ext(required=True) is not None:
+            x = x.contiguous(memory_format=torch.channels_last)
+            return _Conv3x3Fn.apply(x, self.weight)
+        return F.conv2d(x, self.weight.to(x.dtype), padding=1)
+
+    def extra_repr(self):
+        return "{}x{} 3x3 (implicit-GEMM MFMA)".format(self.weight.shape[1],
+                                                       self.weight.shape[0])
+
+
+# ---------------------------------------------------------------------------
 # MFMA GEMM (bf16 inputs, fp32 accumulate) for Dense layers / serving
 # ---------------------------------------------------------------------------
 

@@ -319,3 +319,35 @@ def test_fused_maxpool(dtype):
     tol = (1e-4, 1e-5) if dtype == torch.float32 else (2e-2, 2e-2)
     _close(y, y_ref, *tol, "maxpool y")
     _close(x.grad, x_ref.grad, *tol, "maxpool dx")
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("shape", [(2, 64, 16, 16, 64), (2, 128, 9, 9, 256),
+                                   (1, 32, 14, 15, 512)])
+def test_conv3x3_mfma(shape):
+    """Conv3x3 (implicit-GEMM MFMA) vs F.conv2d: y, dx, dw — padding taps
+    exercise the zero-guard staging path."""
+    from tensorflowonspark_amd.ops.modules import Conv3x3
+    torch.manual_seed(13)
+    N, Cin, H, W, Cout = shape
+    x0 = torch.randn(N, Cin, H, W) / 4
+    wref = torch.randn(Cout, Cin, 3, 3).requires_grad_(True) / 8
+
+    w2 = wref.detach().clone().requires_grad_(True)
+    x_ref = x0.clone().requires_grad_(True)
+    y_ref = torch.nn.functional.conv2d(x_ref, w2, padding=1)
+    gy = torch.randn_like(y_ref) / 4
+    y_ref.backward(gy)
+
+    mod = Conv3x3(Cin, Cout).cuda()
+    with torch.no_grad():
+        mod.weight.copy_(w2.detach())
+    x = x0.cuda().bfloat16().contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    y = mod(x)
+    y.backward(gy.cuda().bfloat16())
+
+    _close(y, y_ref, 3e-2, 6e-2, "conv3x3 y")
+    _close(x.grad, x_ref.grad, 3e-2, 6e-2, "conv3x3 dx")
+    _close(mod.weight.grad, w2.grad, 3e-2, 2e-1, "conv3x3 dw")
