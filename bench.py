@@ -81,12 +81,21 @@ class SparkFeed:
                                    "local")
         from tensorflowonspark_amd.utils import shmring
         ring_name = "tfosr_bench_{}".format(os.getpid())
-        self.ring = shmring.BlockRing(ring_name, 6, slot_bytes,
-                                      data_queue=self.mgr.get_queue("input"),
-                                      free_queue=self.mgr.get_queue("free"),
-                                      create=True)
+        # bound /dev/shm use: 8 ranks x big batches must not exhaust shm
+        slots = max(2, min(6, (1 << 30) // max(1, slot_bytes)))
+        while True:
+            try:
+                self.ring = shmring.BlockRing(
+                    ring_name, slots, slot_bytes,
+                    data_queue=self.mgr.get_queue("input"),
+                    free_queue=self.mgr.get_queue("free"), create=True)
+                break
+            except OSError:
+                if slots <= 2:
+                    raise
+                slots //= 2
         self.mgr.set("ring_name", ring_name)
-        self.mgr.set("ring_slots", 6)
+        self.mgr.set("ring_slots", slots)
         self.mgr.set("ring_slot_bytes", slot_bytes)
         self.mgr.set("state", "running")
         self.stop_evt = multiprocessing.Event()
@@ -180,7 +189,9 @@ def build_model(name, device):
 def main():
     args = parse_args()
     rank = int(os.environ.get("RANK", 0))
-    world = int(os.environ.get("WORLD_SIZE", args.gpus if args.gpus > 1 else 1))
+    # WORLD_SIZE comes from torch.distributed.run; a bare `--gpus N` without
+    # a launcher must NOT wait for peers that will never arrive
+    world = int(os.environ.get("WORLD_SIZE", "1"))
     use_cuda = torch.cuda.is_available()
     distributed = world > 1
 
