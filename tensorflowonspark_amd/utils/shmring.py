@@ -23,18 +23,24 @@ import pickle
 from multiprocessing import shared_memory
 
 
-def _untrack(shm):
-    """Detach a SharedMemory from the resource tracker.
+class _no_track:
+    """Attach SharedMemory without resource-tracker registration.
 
-    Attaching processes are not owners; without this, every attacher's
-    resource_tracker unlinks the segment at interpreter exit (CPython < 3.13
-    has no ``track=False``).
+    Attaching processes are not owners; CPython < 3.13 registers on attach
+    anyway, which makes every attacher's tracker unlink the segment at exit
+    (and later unregisters double-count, spamming KeyError warnings).
+    Suppressing registration during attach avoids both.
     """
-    try:
+
+    def __enter__(self):
         from multiprocessing import resource_tracker
-        resource_tracker.unregister(shm._name, "shared_memory")
-    except Exception:
-        pass
+        self._rt = resource_tracker
+        self._orig = resource_tracker.register
+        resource_tracker.register = lambda *a, **k: None
+        return self
+
+    def __exit__(self, *exc):
+        self._rt.register = self._orig
 
 
 class BlockRing:
@@ -51,8 +57,8 @@ class BlockRing:
             for i in range(slots):
                 free_queue.put(i)
         else:
-            self.shm = shared_memory.SharedMemory(name=name)
-            _untrack(self.shm)
+            with _no_track():
+                self.shm = shared_memory.SharedMemory(name=name)
         self.name = self.shm.name
         self._owner = create
 
