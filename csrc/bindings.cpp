@@ -31,6 +31,8 @@ void tfosr_nhwc_pack(const void*, void*, const float*, const float*, float, int,
                      int, long, int, long, hipStream_t);
 void tfosr_sgd_step(float*, const float*, float*, float, float, float, int, long,
                     hipStream_t);
+void tfosr_adam_step(float*, const float*, float*, float*, float, float, float,
+                     float, float, int, int, long, hipStream_t);
 void tfosr_gemm_bt(const void*, const void*, void*, int, int, int, int,
                    hipStream_t);
 void tfosr_mfma_probe(const short*, const short*, float*, hipStream_t);
@@ -290,6 +292,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("nhwc_pack", &nhwc_pack);
   m.def("sgd_step", &sgd_step);
+  m.def("adam_step", [](at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
+                        double lr, double b1, double b2, double eps, double wd,
+                        long step, bool decoupled) {
+    TORCH_CHECK(p.scalar_type() == at::kFloat);
+    tfosr_adam_step(p.data_ptr<float>(), g.data_ptr<float>(),
+                    m.data_ptr<float>(), v.data_ptr<float>(), lr, b1, b2, eps,
+                    wd, (int)step, decoupled, p.numel(), cur_stream());
+  });
   m.def("gemm_bt", &gemm_bt, py::arg("a"), py::arg("b"), py::arg("out_bf16") = false);
   m.def("gemm_bf16", &gemm_bf16);
   m.def("mfma_probe", &mfma_probe);

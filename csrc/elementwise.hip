@@ -121,3 +121,58 @@ void tfosr_sgd_step(float* p, const float* g, float* m, float lr, float mu,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// Fused flat Adam(W):  m=b1*m+(1-b1)g; v=b2*v+(1-b2)g^2;
+//                      p -= lr * mhat/(sqrt(vhat)+eps) (+ decoupled wd)
+// ---------------------------------------------------------------------------
+
+__global__ void adam_step_kernel(float* __restrict__ p, const float* __restrict__ g,
+                                 float* __restrict__ m, float* __restrict__ v,
+                                 float lr, float b1, float b2, float eps,
+                                 float wd, float bc1, float bc2, int decoupled,
+                                 long n) {
+  constexpr int V = 4;
+  const long nvec = n / V;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    f4v pv = *(f4v*)(p + i * V);
+    f4v gv = *(const f4v*)(g + i * V);
+    f4v mv = *(f4v*)(m + i * V);
+    f4v vv = *(f4v*)(v + i * V);
+    #pragma unroll
+    for (int j = 0; j < V; ++j) {
+      float grad = decoupled ? gv[j] : gv[j] + wd * pv[j];
+      mv[j] = b1 * mv[j] + (1.f - b1) * grad;
+      vv[j] = b2 * vv[j] + (1.f - b2) * grad * grad;
+      float mhat = mv[j] / bc1;
+      float vhat = vv[j] / bc2;
+      float upd = mhat / (sqrtf(vhat) + eps);
+      if (decoupled) upd += wd * pv[j];
+      pv[j] -= lr * upd;
+    }
+    *(f4v*)(m + i * V) = mv;
+    *(f4v*)(v + i * V) = vv;
+    *(f4v*)(p + i * V) = pv;
+  }
+  for (long i = nvec * V + blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float grad = decoupled ? g[i] : g[i] + wd * p[i];
+    m[i] = b1 * m[i] + (1.f - b1) * grad;
+    v[i] = b2 * v[i] + (1.f - b2) * grad * grad;
+    float upd = (m[i] / bc1) / (sqrtf(v[i] / bc2) + eps);
+    if (decoupled) upd += wd * p[i];
+    p[i] -= lr * upd;
+  }
+}
+
+extern "C" void tfosr_adam_step(float* p, const float* g, float* m, float* v,
+                                float lr, float b1, float b2, float eps,
+                                float wd, int step, int decoupled, long n,
+                                hipStream_t s) {
+  int grid = tfosr_grid(n / 4, 256);
+  float bc1 = 1.f - powf(b1, (float)step);
+  float bc2 = 1.f - powf(b2, (float)step);
+  hipLaunchKernelGGL(adam_step_kernel, dim3(grid), dim3(256), 0, s,
+                     p, g, m, v, lr, b1, b2, eps, wd, bc1, bc2, decoupled, n);
+}

@@ -453,3 +453,46 @@ def gemm_bf16(a, b):
         if ext is not None:
             return ext.gemm_bf16(a.contiguous(), b.contiguous())
     return (a.float() @ b.float())
+
+
+class BucketAdam:
+    """Adam(W) over DDPEngine flat buckets — one fused kernel per bucket on
+    GPU (fp32 m/v state), torch ops on CPU. ``decoupled=True`` = AdamW."""
+
+    def __init__(self, engine, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
+                 weight_decay=0.0, decoupled=False):
+        self.engine = engine
+        self.lr = lr
+        self.b1, self.b2 = betas
+        self.eps = eps
+        self.weight_decay = weight_decay
+        self.decoupled = decoupled
+        self.step_count = 0
+        self._m = [torch.zeros_like(b.buffer) for b in engine._buckets]
+        self._v = [torch.zeros_like(b.buffer) for b in engine._buckets]
+
+    @torch.no_grad()
+    def step(self):
+        self.step_count += 1
+        for bucket, m, v in zip(self.engine._buckets, self._m, self._v):
+            pf = bucket.param_flat
+            g = bucket.buffer
+            assert pf is not None, "BucketAdam requires flatten_params=True"
+            ext = get_ext(required=True) if g.is_cuda else None
+            if ext is not None:
+                ext.adam_step(pf, g, m, v, self.lr, self.b1, self.b2, self.eps,
+                              self.weight_decay, self.step_count, self.decoupled)
+                continue
+            grad = g if self.decoupled or not self.weight_decay \
+                else g.add(pf, alpha=self.weight_decay)
+            m.mul_(self.b1).add_(grad, alpha=1 - self.b1)
+            v.mul_(self.b2).addcmul_(grad, grad, value=1 - self.b2)
+            bc1 = 1 - self.b1 ** self.step_count
+            bc2 = 1 - self.b2 ** self.step_count
+            upd = (m / bc1) / ((v / bc2).sqrt() + self.eps)
+            if self.decoupled and self.weight_decay:
+                upd = upd + self.weight_decay * pf
+            pf.add_(upd, alpha=-self.lr)
+
+    def zero_grad(self):
+        self.engine.zero_grad()

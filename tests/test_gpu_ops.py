@@ -351,3 +351,25 @@ def test_conv3x3_mfma(shape):
     _close(y, y_ref, 3e-2, 6e-2, "conv3x3 y")
     _close(x.grad, x_ref.grad, 3e-2, 6e-2, "conv3x3 dx")
     _close(mod.weight.grad, w2.grad, 3e-2, 2e-1, "conv3x3 dw")
+
+
+@gpu
+@requires_gpu
+def test_adam_step():
+    ext = _ext()
+    torch.manual_seed(14)
+    n = 4099
+    p = torch.randn(n, device="cuda")
+    g = torch.randn(n, device="cuda")
+    m = torch.zeros(n, device="cuda")
+    v = torch.zeros(n, device="cuda")
+    p_ref = p.clone()
+
+    # two fused steps vs torch.optim.Adam on the same grads
+    ref_p = p_ref.clone().requires_grad_(False)
+    ref = torch.optim.Adam([ref_p], lr=0.01, betas=(0.9, 0.999), eps=1e-8)
+    for step in (1, 2):
+        ext.adam_step(p, g, m, v, 0.01, 0.9, 0.999, 1e-8, 0.0, step, False)
+        ref_p.grad = g.clone()
+        ref.step()
+    _close(p, ref_p, 1e-5, 1e-6, "adam p")
