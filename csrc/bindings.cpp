@@ -42,6 +42,8 @@ void tfosr_maxpool_bwd(const void*, const unsigned char*, void*, int, int, int,
                        int, int, int, int, int, int, int, hipStream_t);
 void tfosr_conv3x3(const void*, const void*, const void*, void*, int, int, int,
                    int, int, int, int, int, int, int, hipStream_t);
+void tfosr_conv_wrw(const void*, const void*, float*, int, int, int, int, int,
+                    int, int, int, int, int, hipStream_t);
 }
 
 namespace tfosr {
@@ -334,6 +336,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                   y.data_ptr(), /*out_bf16=*/1, N, H, W, Cin, Cout, OH, OW,
                   S, P, cur_stream());
     return y;
+  });
+  m.def("conv_wrw", [](at::Tensor dy, at::Tensor x, long R, long S, long P) {
+    TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                x.is_contiguous(at::MemoryFormat::ChannelsLast));
+    TORCH_CHECK(dy.scalar_type() == at::kBFloat16 &&
+                x.scalar_type() == at::kBFloat16);
+    int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+    int Cout = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+    auto dW = at::zeros({Cout, R * S * (long)Cin},
+                        x.options().dtype(at::kFloat));
+    tfosr_conv_wrw(dy.data_ptr(), x.data_ptr(), dW.data_ptr<float>(), N, H, W,
+                   Cin, Cout, OH, OW, R, S, P, cur_stream());
+    return dW;
   });
   m.def("maxpool_bwd", [](at::Tensor dy, at::Tensor idx, long H, long W,
                           long K, long S, long P) {

@@ -327,14 +327,20 @@ class _Conv1x1Fn(torch.autograd.Function):
         # dx[M,Cin] = dy[M,Cout] @ W[Cout,Cin]  ->  gemm_bt(dy, W^T)
         dx2d = ext.gemm_bt(dy2d, w2d.t().contiguous(), True)
         dx = dx2d.view(N, H, W, Cin).permute(0, 3, 1, 2)
-        # dW[Cout,Cin] = dy^T @ x: a reduction over the huge M dim. MIOpen's
-        # tuned wrw igemm beats hipBLASLt's split-K pick here by ~7x, so route
-        # through convolution_backward (weight-only) on the 4-D views.
+        # dW[Cout,Cin] = dy^T @ x: a reduction over the huge M dim.
+        # TFOS_WRW=mfma -> our TN kernel; default MIOpen wrw igemm
+        # (hipBLASLt's split-K pick was ~7x slower than either).
+        import os
         x4d = x2d.view(N, H, W, Cin).permute(0, 3, 1, 2)
-        w4d = w2d.view(Cout, Cin, 1, 1)
-        _, dw, _ = torch.ops.aten.convolution_backward(
-            dy, x4d, w4d, None, [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
-            [False, True, False])
+        if os.environ.get("TFOS_WRW", "mfma") == "mfma" and Cin % 8 == 0 \
+                and Cout % 8 == 0:
+            ext2 = get_ext(required=True)
+            dw = ext2.conv_wrw(dy, x4d, 1, 1, 0).view(Cout, Cin, 1, 1)
+        else:
+            w4d = w2d.view(Cout, Cin, 1, 1)
+            _, dw, _ = torch.ops.aten.convolution_backward(
+                dy, x4d, w4d, None, [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+                [False, True, False])
         return dx, dw.to(wdtype)
 
 
@@ -404,11 +410,17 @@ class _Conv3x3Fn(torch.autograd.Function):
         # W'[cin][r][s][cout] = W[cout][2-r][2-s][cin]
         w9p = weight.flip(2, 3).permute(1, 2, 3, 0)             .reshape(Cin, 9 * Cout).to(torch.bfloat16).contiguous()
         dx = ext.conv3x3_fwd(dy, w9p, Cin, 1, 1)
-        w4 = weight.to(torch.bfloat16).contiguous(
-            memory_format=torch.channels_last)
-        _, dw, _ = torch.ops.aten.convolution_backward(
-            dy, x, w4, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
-            [False, True, False])
+        import os
+        if os.environ.get("TFOS_WRW", "mfma") == "mfma":
+            # dW9[cout][r][s][cin] -> [Cout, Cin, 3, 3]
+            dw9 = ext.conv_wrw(dy, x, 3, 3, 1)
+            dw = dw9.view(Cout, 3, 3, Cin).permute(0, 3, 1, 2).contiguous()
+        else:
+            w4 = weight.to(torch.bfloat16).contiguous(
+                memory_format=torch.channels_last)
+            _, dw, _ = torch.ops.aten.convolution_backward(
+                dy, x, w4, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+                [False, True, False])
         return dx, dw.to(weight.dtype)
 
 
