@@ -71,11 +71,11 @@ __global__ __launch_bounds__(256, 2) void conv_wrw_kernel(
   const long mb1 = (mb0 + per < mblocks) ? (mb0 + per) : mblocks;
   if (mb0 >= mb1) return;
 
-  // this thread's staged chunks: 2 per operand tile
-  // chunk over 64 m-rows x 16 col-chunks (8 cols each)
-  int cm[2], cc[2];
+  // this thread's staged chunks: 64 m-rows x 16 col-chunks (8 cols each)
+  // = 1024 chunks per operand tile -> 4 per thread
+  int cm[4], cc[4];
   #pragma unroll
-  for (int u = 0; u < 2; ++u) {
+  for (int u = 0; u < 4; ++u) {
     int chunk = t + u * 256;
     cm[u] = chunk >> 4;            // m row within block (0..63)
     cc[u] = (chunk & 15) * 8;      // col0 (0..120)
@@ -83,7 +83,7 @@ __global__ __launch_bounds__(256, 2) void conv_wrw_kernel(
 
   auto load_regs = [&](long mb, bf16x8* rdy, bf16x8* rx) {
     #pragma unroll
-    for (int u = 0; u < 2; ++u) {
+    for (int u = 0; u < 4; ++u) {
       long m = mb * WW_BK + cm[u];
       bf16x8 zero = {0, 0, 0, 0, 0, 0, 0, 0};
       rdy[u] = zero;
@@ -108,7 +108,7 @@ __global__ __launch_bounds__(256, 2) void conv_wrw_kernel(
     __attribute__((address_space(3))) char* ld = lds3 + buf * 2 * WW_TILE;
     __attribute__((address_space(3))) char* lx = ld + WW_TILE;
     #pragma unroll
-    for (int u = 0; u < 2; ++u) {
+    for (int u = 0; u < 4; ++u) {
       ww_write8(ld, cc[u], cm[u], rdy[u]);
       ww_write8(lx, cc[u], cm[u], rx[u]);
     }
@@ -120,7 +120,7 @@ __global__ __launch_bounds__(256, 2) void conv_wrw_kernel(
     #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  bf16x8 rdyA[2], rxA[2], rdyB[2], rxB[2];
+  bf16x8 rdyA[4], rxA[4], rdyB[4], rxB[4];
   load_regs(mb0, rdyA, rxA);
   write_lds(0, rdyA, rxA);
 
