@@ -332,7 +332,7 @@ class _Conv1x1Fn(torch.autograd.Function):
         # (hipBLASLt's split-K pick was ~7x slower than either).
         import os
         x4d = x2d.view(N, H, W, Cin).permute(0, 3, 1, 2)
-        if os.environ.get("TFOS_WRW", "mfma") == "mfma" and Cin % 8 == 0 \
+        if os.environ.get("TFOS_WRW", "miopen") == "mfma" and Cin % 8 == 0 \
                 and Cout % 8 == 0:
             ext2 = get_ext(required=True)
             dw = ext2.conv_wrw(dy, x4d, 1, 1, 0).view(Cout, Cin, 1, 1)
@@ -411,7 +411,7 @@ class _Conv3x3Fn(torch.autograd.Function):
         w9p = weight.flip(2, 3).permute(1, 2, 3, 0)             .reshape(Cin, 9 * Cout).to(torch.bfloat16).contiguous()
         dx = ext.conv3x3_fwd(dy, w9p, Cin, 1, 1)
         import os
-        if os.environ.get("TFOS_WRW", "mfma") == "mfma":
+        if os.environ.get("TFOS_WRW", "miopen") == "mfma":
             # dW9[cout][r][s][cin] -> [Cout, Cin, 3, 3]
             dw9 = ext.conv_wrw(dy, x, 3, 3, 1)
             dw = dw9.view(Cout, 3, 3, Cin).permute(0, 3, 1, 2).contiguous()

@@ -373,3 +373,32 @@ def test_adam_step():
         ref_p.grad = g.clone()
         ref.step()
     _close(p, ref_p, 1e-5, 1e-6, "adam p")
+
+
+@gpu
+@requires_gpu
+def test_conv_wrw_kernel():
+    """TN weight-grad kernel (ext.conv_wrw) vs autograd reference — kept as a
+    tested alternative backend (TFOS_WRW=mfma); MIOpen's wrw is the default
+    for speed."""
+    ext = _ext()
+    torch.manual_seed(15)
+    N, Cin, H, W, Cout = 2, 64, 10, 11, 128
+    x = (torch.randn(N, Cin, H, W, device="cuda") / 4).bfloat16() \
+        .contiguous(memory_format=torch.channels_last)
+    dy = (torch.randn(N, Cout, H, W, device="cuda") / 4).bfloat16() \
+        .contiguous(memory_format=torch.channels_last)
+    dw9 = ext.conv_wrw(dy, x, 3, 3, 1)
+    dw = dw9.view(Cout, 3, 3, Cin).permute(0, 3, 1, 2)
+
+    w = torch.zeros(Cout, Cin, 3, 3, device="cuda", requires_grad=True)
+    y = torch.nn.functional.conv2d(x.float(), w, padding=1)
+    y.backward(dy.float())
+    _close(dw, w.grad, 3e-2, 2e-1, "conv_wrw dw")
+
+    # 1x1 variant
+    dw1 = ext.conv_wrw(dy, x.narrow(1, 0, Cin), 1, 1, 0).view(Cout, Cin, 1, 1)
+    w1 = torch.zeros(Cout, Cin, 1, 1, device="cuda", requires_grad=True)
+    y1 = torch.nn.functional.conv2d(x.float(), w1)
+    y1.backward(dy.float())
+    _close(dw1, w1.grad, 3e-2, 2e-1, "conv_wrw 1x1 dw")
