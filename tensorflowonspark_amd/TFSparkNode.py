@@ -30,7 +30,7 @@ import time
 import traceback
 import uuid
 
-from . import TFManager, gpu_info, marker, reservation, util
+from . import TFManager, gpu_info, reservation, util
 
 logger = logging.getLogger(__name__)
 
