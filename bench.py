@@ -82,7 +82,8 @@ class SparkFeed:
         from tensorflowonspark_amd.utils import shmring
         ring_name = "tfosr_bench_{}".format(os.getpid())
         # bound /dev/shm use: 8 ranks x big batches must not exhaust shm
-        slots = max(2, min(6, (1 << 30) // max(1, slot_bytes)))
+        # 4 slots x ~150MB x 8 ranks stays well inside /dev/shm
+        slots = max(2, min(4, (1 << 30) // max(1, slot_bytes)))
         while True:
             try:
                 self.ring = shmring.BlockRing(
