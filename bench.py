@@ -33,7 +33,8 @@ def parse_args():
     p.add_argument("--batch", type=int, default=0,
                    help="per-GPU batch size (0 = auto: 1024 on GPU, 8 on CPU)")
     p.add_argument("--model", default="resnet50",
-                   choices=["resnet50", "resnet56_cifar", "mnist_cnn", "unet", "deeplabv3"])
+                   choices=["resnet50", "resnet56_cifar", "mnist_cnn", "mnist_mlp",
+                            "unet", "deeplabv3"])
     p.add_argument("--feed", default="spark", choices=["spark", "device"],
                    help="spark = real shm-ring ingest path; device = on-GPU synthetic")
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
@@ -178,6 +179,9 @@ def build_model(name, device):
         return models.resnet56_cifar().to(device), (32, 32, 3), 10
     if name == "mnist_cnn":
         return models.MNISTNet().to(device), (28, 28, 1), 10
+    if name == "mnist_mlp":
+        # BASELINE config 1: world_size=2 CPU/gloo plumbing model
+        return models.MNISTMLP().to(device), (28, 28, 1), 10
     if name == "unet":
         from tensorflowonspark_amd.models.segmentation import unet_mobilenet
         return unet_mobilenet(num_classes=3).to(device), (128, 128, 3), 3
