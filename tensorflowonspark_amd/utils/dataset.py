@@ -59,7 +59,7 @@ class TFRecordDataset:
     """
 
     def __init__(self, path, shard_index=0, num_shards=1, batch_size=1,
-                 decode=True, drop_remainder=False):
+                 decode=True, drop_remainder=False, shuffle_buffer=0, seed=0):
         if path.startswith("file://"):
             path = path[len("file://"):]
         self.files = shard_files(path, shard_index, num_shards)
@@ -70,10 +70,29 @@ class TFRecordDataset:
         self.batch_size = batch_size
         self.decode = decode
         self.drop_remainder = drop_remainder
+        self.shuffle_buffer = shuffle_buffer
+        self.seed = seed
+        self._epoch = 0
+
+    def _records(self):
+        files = list(self.files)
+        if self.shuffle_buffer:
+            import random
+            rng = random.Random((self.seed, self._epoch))
+            rng.shuffle(files)
+            buf = []
+            for ex in tfrecord_examples(files, self.decode):
+                buf.append(ex)
+                if len(buf) >= self.shuffle_buffer:
+                    yield buf.pop(rng.randrange(len(buf)))
+            while buf:
+                yield buf.pop(rng.randrange(len(buf)))
+        else:
+            yield from tfrecord_examples(files, self.decode)
 
     def __iter__(self):
-        return batched(tfrecord_examples(self.files, self.decode),
-                       self.batch_size, self.drop_remainder)
+        self._epoch += 1  # new shuffle order per epoch
+        return batched(self._records(), self.batch_size, self.drop_remainder)
 
     def count(self):
         return sum(1 for _ in tfrecord_examples(self.files, decode=False))

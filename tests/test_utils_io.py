@@ -120,3 +120,14 @@ def test_cluster_tensorflow_mode(tmp_path):
             assert abs(w - 2.0) < 0.1, weights
     finally:
         sc.stop()
+
+
+def test_tfrecord_dataset_shuffle(tmp_path):
+    total = _write_tfrecords(str(tmp_path), nfiles=2, per_file=30)
+    ds = TFRecordDataset(str(tmp_path), 0, 1, batch_size=60, shuffle_buffer=16)
+    epoch1 = [ex["x"][1][0] for b in ds for ex in b]
+    epoch2 = [ex["x"][1][0] for b in ds for ex in b]
+    assert len(epoch1) == len(epoch2) == total
+    assert sorted(epoch1) == sorted(epoch2)       # same elements
+    assert epoch1 != sorted(epoch1)               # actually shuffled
+    assert epoch1 != epoch2                       # reshuffled per epoch
