@@ -114,18 +114,23 @@ class TFCluster(object):
                         ssc.stop(stopSparkContext=False, stopGraceFully=True)
 
             if self.input_mode == InputMode.TENSORFLOW:
-                # wait for worker-role tasks to complete on their own
-                # (workers read data directly; the start job finishes when the
-                # user fn returns — poll the background thread's liveness)
-                while self._start_thread.is_alive() and not tf_status.get("error"):
+                # wait for all *worker*-role map_funs to finish (they read
+                # data directly); ps/evaluator bootstrap tasks stay alive by
+                # design until we stop them below — the reference polled
+                # statusTracker for the same condition (TFCluster.py:154-169)
+                target = len(workers)
+                while (self.server.done_count() < target
+                       and self._start_thread.is_alive()
+                       and not tf_status.get("error")):
                     time.sleep(1)
 
             # push end-of-feed into worker queues
-            workerRDD = self.sc.parallelize(
-                [n["executor_id"] for n in workers], len(workers))
-            workerRDD.foreachPartition(
-                TFSparkNode.shutdown(self.cluster_info, self.queues,
-                                     grace_secs=grace_secs))
+            if workers:
+                workerRDD = self.sc.parallelize(
+                    [n["executor_id"] for n in workers], len(workers))
+                workerRDD.foreachPartition(
+                    TFSparkNode.shutdown(self.cluster_info, self.queues,
+                                         grace_secs=grace_secs))
 
             if tf_status.get("error"):
                 logger.error("cluster error: %s", tf_status["error"])

@@ -466,6 +466,15 @@ def run(fn, tf_args, cluster_meta, tensorboard=False, log_dir=None, queues=None,
             except Exception:
                 mgr.get_queue("error").put(traceback.format_exc())
                 raise
+            finally:
+                # signal completion so shutdown() can distinguish "workers
+                # finished" from "ps/evaluator still serving"
+                try:
+                    done_client = reservation.Client(cluster_meta["server_addr"])
+                    done_client.notify_done(executor_id)
+                    done_client.close()
+                except Exception as e:
+                    logger.debug("completion notify failed: %s", e)
             logger.info("%s node %d completed", job_name, task_index)
 
         return []

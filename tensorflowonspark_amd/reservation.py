@@ -104,6 +104,8 @@ class Server(MessageSocket):
         self.done = threading.Event()
         self._listener = None
         self._thread = None
+        self._done_nodes = []
+        self._done_lock = threading.Lock()
 
     # -- lifecycle -----------------------------------------------------------
 
@@ -140,6 +142,10 @@ class Server(MessageSocket):
         self._thread = threading.Thread(target=self._serve, daemon=True)
         self._thread.start()
         return addr
+
+    def done_count(self):
+        with self._done_lock:
+            return len(self._done_nodes)
 
     def stop(self):
         self.done.set()
@@ -217,6 +223,13 @@ class Server(MessageSocket):
             self.send(sock, {"type": "RESP", "data": self.reservations.done()})
         elif mtype == "QINFO":
             self.send(sock, {"type": "RESP", "data": self.reservations.get()})
+        elif mtype == "DONE":
+            # a worker's map_fun returned (InputMode.TENSORFLOW completion
+            # signal — the analog of the reference's statusTracker polling,
+            # reference TFCluster.py:154-169)
+            with self._done_lock:
+                self._done_nodes.append(msg.get("data"))
+            self.send(sock, {"type": "OK"})
         elif mtype == "STOP":
             self.reservations.stop()
             self.send(sock, {"type": "OK"})
@@ -282,6 +295,10 @@ class Client(MessageSocket):
 
     def request_stop(self):
         return self._request({"type": "STOP"})
+
+    def notify_done(self, executor_id):
+        """Report this worker's map_fun completion (TENSORFLOW mode)."""
+        return self._request({"type": "DONE", "data": executor_id})
 
 
 def _parse_port_spec(spec):

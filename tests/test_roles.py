@@ -94,3 +94,31 @@ def test_driver_ps_nodes():
             assert last < first
     finally:
         sc.stop()
+
+
+def _tf_mode_ps_fn(args, ctx):
+    if ctx.job_name == "ps":
+        ctx.run_parameter_server()
+        return
+    # TENSORFLOW-mode worker: no feed, just signal it ran
+    with open("tfmode_worker.txt", "w") as f:
+        f.write("ran")
+
+
+@pytest.mark.timeout(300)
+def test_tensorflow_mode_with_ps_shuts_down():
+    """shutdown() must not wait forever on ps bootstrap tasks when workers
+    read data directly (InputMode.TENSORFLOW completion signaling)."""
+    sc = LocalSparkContext(num_executors=3)
+    try:
+        cluster = TFCluster.run(sc, _tf_mode_ps_fn, {}, num_executors=3,
+                                num_ps=1, master_node=None,
+                                input_mode=TFCluster.InputMode.TENSORFLOW,
+                                num_gpus=0, reservation_timeout=60)
+        cluster.shutdown(grace_secs=0)
+        import glob
+        hits = glob.glob(os.path.join(sc._root, "executor_*",
+                                      "tfmode_worker.txt"))
+        assert len(hits) == 2
+    finally:
+        sc.stop()
