@@ -217,14 +217,15 @@ def _has_spark_resource_api():
         return False
 
 
-def _get_gpus(num_gpus, worker_index=-1, cluster_spec=None, my_host=None):
+def _get_gpus(num_gpus, worker_index=-1):
     """Resolve this node's GPU assignment, in priority order:
-    (1) Spark 3 TaskContext resources, (2) K8s pod (all visible), (3)
-    rocm-smi probing with same-host peer index (reference ``TFSparkNode.py:179-239``).
-    Returns a comma string for HIP_VISIBLE_DEVICES, or None for CPU."""
+    (1) Spark 3 TaskContext resources, (2) K8s pod (granted by the scheduler),
+    (3) rocm-smi/amd-smi probing (reference ``TFSparkNode.py:179-239``).
+    Returns (gpu_string_or_None, from_probe): ``from_probe`` marks assignments
+    that must be re-derived once the roster gives this node its deterministic
+    index among same-host peers (second pass, reference ``:386-388``)."""
     if num_gpus == 0:
-        return None
-    gpus = None
+        return None, False
     if _has_spark_resource_api():
         try:
             from pyspark import TaskContext
@@ -232,27 +233,28 @@ def _get_gpus(num_gpus, worker_index=-1, cluster_spec=None, my_host=None):
             if tc is not None:
                 resources = tc.resources()
                 if resources and "gpu" in resources:
-                    gpus = ",".join(resources["gpu"].addresses[:num_gpus])
+                    return ",".join(resources["gpu"].addresses[:num_gpus]), False
         except Exception:
-            gpus = None
-    if gpus is None and os.environ.get("SPARK_EXECUTOR_POD_IP"):
+            pass
+    if os.environ.get("SPARK_EXECUTOR_POD_IP"):
         # K8s: the pod was granted its GPUs by the scheduler; use them all
         ids = gpu_info._list_gpu_ids()
-        gpus = ",".join(str(i) for i in ids[:num_gpus]) if ids else None
-    if gpus is None and gpu_info.is_gpu_available():
-        # index among peers on the same host so co-located workers get
-        # disjoint GPUs (reference allocation-index math, TFSparkNode.py:213-228)
-        idx = -1
-        if cluster_spec and my_host:
-            peers = sorted(addr for addrs in cluster_spec.values() for addr in addrs
-                           if addr.rsplit(":", 1)[0] == my_host)
-            mine = [a for a in peers]
-            for i, a in enumerate(mine):
-                if my_host and a.rsplit(":", 1)[0] == my_host:
-                    idx = i
-                    break
-        gpus = gpu_info.get_gpus(num_gpus, idx, format=str)
-    return gpus
+        if ids:
+            return ",".join(str(i) for i in ids[:num_gpus]), False
+        return None, False
+    if gpu_info.is_gpu_available():
+        return gpu_info.get_gpus(num_gpus, worker_index, format=str), True
+    return None, False
+
+
+def _host_peer_index(cluster_info, host, executor_id):
+    """This node's rank among GPU-wanting nodes on the same host — the
+    deterministic slice index for ``gpu_info.get_gpus`` (co-located executors
+    must get disjoint GPUs; reference allocation-index math,
+    ``TFSparkNode.py:213-228``)."""
+    peers = sorted(n["executor_id"] for n in cluster_info
+                   if n["host"] == host)
+    return peers.index(executor_id)
 
 
 def _get_cluster_spec(cluster_info):
@@ -287,7 +289,7 @@ def run(fn, tf_args, cluster_meta, tensorboard=False, log_dir=None, queues=None,
         # -- GPU probe & pinning ---------------------------------------------
         num_gpus = int(cluster_meta.get("num_gpus", 1))
         wants_gpu = num_gpus > 0 and job_name in ("worker", "chief", "master", "evaluator")
-        gpu_str = _get_gpus(num_gpus if wants_gpu else 0)
+        gpu_str, gpu_from_probe = _get_gpus(num_gpus if wants_gpu else 0)
         if gpu_str is not None:
             os.environ["HIP_VISIBLE_DEVICES"] = gpu_str
             os.environ["CUDA_VISIBLE_DEVICES"] = gpu_str
@@ -402,6 +404,17 @@ def run(fn, tf_args, cluster_meta, tensorboard=False, log_dir=None, queues=None,
         client.close()
 
         cluster_spec = _get_cluster_spec(cluster_info)
+
+        # second GPU pass: with the roster known, probe-based assignments
+        # become deterministic slices so co-located executors get disjoint
+        # GPUs (reference TFSparkNode.py:386-388)
+        if gpu_from_probe and wants_gpu:
+            idx = _host_peer_index(cluster_info, host, executor_id)
+            gpu_str = gpu_info.get_gpus(num_gpus, idx, format=str)
+            os.environ["HIP_VISIBLE_DEVICES"] = gpu_str
+            os.environ["CUDA_VISIBLE_DEVICES"] = gpu_str
+            logger.info("executor %d re-pinned to GPU(s) %s (host index %d)",
+                        executor_id, gpu_str, idx)
         # TF_CONFIG-equivalent for tooling that wants it
         os.environ["TFOS_CLUSTER_SPEC"] = json.dumps(
             {"cluster": cluster_spec,

@@ -103,3 +103,14 @@ def test_hdfs_path_matrix():
     assert p.startswith("hdfs://nn:8020/user/") and p.endswith("/rel")
     # relative on local fs -> cwd
     assert TFNode.hdfs_path(_Ctx("file://", "/tmp/wd"), "rel") == "file:///tmp/wd/rel"
+
+
+def test_host_peer_index():
+    from tensorflowonspark_amd.TFSparkNode import _host_peer_index
+    info = [{"host": "h1", "executor_id": 3}, {"host": "h1", "executor_id": 0},
+            {"host": "h2", "executor_id": 1}, {"host": "h1", "executor_id": 5}]
+    # sorted peers on h1: [0, 3, 5] -> deterministic disjoint slice indices
+    assert _host_peer_index(info, "h1", 0) == 0
+    assert _host_peer_index(info, "h1", 3) == 1
+    assert _host_peer_index(info, "h1", 5) == 2
+    assert _host_peer_index(info, "h2", 1) == 0
