@@ -21,9 +21,9 @@ def train_fn(args, ctx):
     engine = DDPEngine(model, bucket_mb=4)
     opt = BucketSGD(engine, lr=0.05, momentum=0.9)
     feed = ctx.get_data_feed(train_mode=True)
-    while not feed.should_stop():
-        batch = feed.next_batch(args.batch_size)
-        if not batch:
+    while True:
+        batch = feed.next_batch(args.batch_size) if not feed.should_stop() else []
+        if not engine.all_ranks_ready(len(batch) > 0):
             break
         import numpy as np
         img = np.asarray([r[0] for r in batch], dtype=np.float32) / 255.0

@@ -40,9 +40,10 @@ def map_fun(args, ctx):
     model.train()
 
     step = start_step
-    while not feed.should_stop():
-        batch = feed.next_batch(args.batch_size)
-        if not batch:
+    while True:
+        batch = feed.next_batch(args.batch_size) if not feed.should_stop() else []
+        # built-in uneven-partition guard: stop together once any rank runs dry
+        if not engine.all_ranks_ready(len(batch) > 0):
             break
         import numpy as np
         arr = np.asarray(batch, dtype=np.float32)

@@ -48,9 +48,9 @@ def map_fun(args, ctx):
     amp = torch.autocast(device.type, dtype=torch.bfloat16, enabled=use_cuda)
 
     step = 0
-    while not feed.should_stop():
-        batch = feed.next_batch(args.batch_size)
-        if not batch:
+    while True:
+        batch = feed.next_batch(args.batch_size) if not feed.should_stop() else []
+        if not engine.all_ranks_ready(len(batch) > 0):
             break
         imgs = np.asarray([r[0] for r in batch], dtype=np.uint8).reshape(
             (-1,) + shape)

@@ -33,9 +33,9 @@ def map_fun(args, ctx):
     feed = ctx.get_data_feed(train_mode=True)
     amp = torch.autocast(device.type, dtype=torch.bfloat16, enabled=use_cuda)
 
-    while not feed.should_stop():
-        batch = feed.next_batch(args.batch_size)
-        if not batch:
+    while True:
+        batch = feed.next_batch(args.batch_size) if not feed.should_stop() else []
+        if not engine.all_ranks_ready(len(batch) > 0):
             break
         imgs = np.asarray([r[0] for r in batch], dtype=np.uint8).reshape(
             -1, 128, 128, 3)

@@ -161,6 +161,23 @@ class DDPEngine:
                 if p.grad is None or p.grad.data_ptr() != bucket.views[p].data_ptr():
                     p.grad = bucket.views[p]
 
+    def all_ranks_ready(self, have_data):
+        """Uneven-partition guard, built in (the reference forced every user
+        to hand-code a 90%-of-steps workaround against sync all-reduce hangs,
+        reference ``mnist_spark.py:58-64``): returns True only while EVERY
+        rank still has data. Call once per step with ``len(batch) > 0`` and
+        break when False — at most one partial step per rank is dropped, and
+        no rank ever blocks in an all-reduce its peers will never enter."""
+        if not self.enabled:
+            return bool(have_data)
+        device = self._buckets[0].buffer.device if self._buckets else None
+        flag = torch.tensor([1 if have_data else 0],
+                            dtype=torch.int32,
+                            device=device if device is not None
+                            and device.type == "cuda" else "cpu")
+        dist.all_reduce(flag, op=dist.ReduceOp.MIN, group=self.pg)
+        return bool(flag.item())
+
     class _NoSync:
         def __init__(self, engine):
             self.engine = engine

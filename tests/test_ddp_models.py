@@ -101,3 +101,51 @@ def test_world2_gloo_params_stay_in_sync():
         assert p.exitcode == 0
     assert results[0] == pytest.approx(results[1]), \
         "ranks diverged: {} vs {}".format(results[0], results[1])
+
+
+def _uneven_worker(rank, world, port, result_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    model = MNISTMLP(hidden=16)
+    engine = DDPEngine(model, bucket_mb=1)
+    opt = BucketSGD(engine, lr=0.05)
+    # rank 0 has 5 batches, rank 1 only 3: without the guard rank 0's 4th
+    # all-reduce would hang forever
+    nbatches = 5 if rank == 0 else 3
+    data = [torch.randn(8, 784) for _ in range(nbatches)]
+    steps = 0
+    it = iter(data)
+    while True:
+        batch = next(it, None)
+        if not engine.all_ranks_ready(batch is not None):
+            break
+        opt.zero_grad()
+        loss = softmax_cross_entropy(model(batch), torch.randint(0, 10, (8,)))
+        loss.backward()
+        engine.finalize_backward()
+        opt.step()
+        steps += 1
+    result_q.put((rank, steps))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_uneven_partition_guard():
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    procs = [ctx.Process(target=_uneven_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = dict(q.get(timeout=120) for _ in range(2))
+    for p in procs:
+        p.join(timeout=30)
+        assert p.exitcode == 0
+    # both ranks stopped together after the shorter feed (3 steps)
+    assert results == {0: 3, 1: 3}

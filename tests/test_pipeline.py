@@ -44,9 +44,9 @@ def _train_fn(args, ctx):
     engine = DDPEngine(model, bucket_mb=1)
     opt = BucketSGD(engine, lr=0.2, momentum=0.0)
     feed = ctx.get_data_feed(train_mode=True)
-    while not feed.should_stop():
-        batch = feed.next_batch(args.batch_size)
-        if not batch:
+    while True:
+        batch = feed.next_batch(args.batch_size) if not feed.should_stop() else []
+        if not engine.all_ranks_ready(len(batch) > 0):
             break
         x = torch.tensor([r[0] for r in batch], dtype=torch.float32)
         y = torch.tensor([[r[1]] for r in batch], dtype=torch.float32)
