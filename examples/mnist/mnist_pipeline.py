@@ -35,6 +35,7 @@ def train_fn(args, ctx):
         loss.backward()
         engine.finalize_backward()
         opt.step()
+    feed.terminate()
     if ctx.is_chief:
         ctx.export_saved_model(model.cpu(), args.export_dir)
     torch.distributed.destroy_process_group()

@@ -60,6 +60,9 @@ def map_fun(args, ctx):
         if step % 200 == 0 and ctx.is_chief:
             ckpt.save_checkpoint(args.model_dir, step, model)
 
+    # drain anything this rank didn't consume (reference mnist_spark.py:71 —
+    # leftover queued blocks would otherwise trip the feeder's feed_timeout)
+    feed.terminate()
     if ctx.is_chief:
         ckpt.save_checkpoint(args.model_dir, step, model)
         ctx.export_saved_model(model.cpu(), args.export_dir)

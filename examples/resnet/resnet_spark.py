@@ -67,6 +67,7 @@ def map_fun(args, ctx):
         opt.step()
         timer.step()
         step += 1
+    feed.terminate()
     if ctx.is_chief:
         ckpt.save_checkpoint(args.model_dir, step, model)
     torch.distributed.destroy_process_group()

@@ -54,6 +54,7 @@ def map_fun(args, ctx):
         loss.backward()
         engine.finalize_backward()
         opt.step()
+    feed.terminate()
     if ctx.is_chief:
         ctx.export_saved_model(model.cpu(), args.export_dir)
     torch.distributed.destroy_process_group()

@@ -97,3 +97,50 @@ def test_error_after_feeding_caught_by_shutdown(sc):
     cluster.train(rdd, num_epochs=1)
     with pytest.raises(Exception, match="late failure"):
         cluster.shutdown(grace_secs=3)
+
+
+def _guarded_train_fn(args, ctx):
+    from tensorflowonspark_amd.parallel import DDPEngine
+    import torch
+    ctx.init_process_group(backend="gloo")
+    model = torch.nn.Linear(2, 1)
+    engine = DDPEngine(model, bucket_mb=1)
+    steps = 0
+    feed = ctx.get_data_feed(train_mode=True)
+    while True:
+        batch = feed.next_batch(40) if not feed.should_stop() else []
+        if not engine.all_ranks_ready(len(batch) > 0):
+            break
+        steps += 1
+    feed.terminate()  # drain unconsumed blocks so the feeder's join returns
+    with open("guard_steps.txt", "w") as f:
+        f.write(str(steps))
+    torch.distributed.destroy_process_group()
+
+
+def test_uneven_feed_through_cluster(sc):
+    """3 partitions over 2 workers: one rank gets ~2x the data; the guard must
+    stop both together and terminate() must unblock the feeder."""
+    cluster = TFCluster.run(sc, _guarded_train_fn, {}, num_executors=2,
+                            num_ps=0, master_node="chief",
+                            input_mode=TFCluster.InputMode.SPARK,
+                            num_gpus=0, reservation_timeout=60)
+    rdd = sc.parallelize(range(600), 3)
+    cluster.train(rdd, num_epochs=1, feed_timeout=60)
+    cluster.shutdown(grace_secs=1)
+    import glob
+    import os
+    import time
+    # workers finish terminate()'s drain (5 s empty-queue timeout) after
+    # shutdown returns; poll for their results
+    deadline = time.time() + 30
+    steps = []
+    while time.time() < deadline:
+        steps = [int(open(f).read()) for f in
+                 glob.glob(os.path.join(sc._root, "executor_*",
+                                        "guard_steps.txt"))]
+        if len(steps) == 2:
+            break
+        time.sleep(0.5)
+    assert len(steps) == 2
+    assert steps[0] == steps[1], steps  # ranks stopped together
