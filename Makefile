@@ -1,0 +1,21 @@
+# Build/test entry points (CI shape parity with the reference's tox targets).
+
+.PHONY: build test test-gpu bench lint clean
+
+build:
+	python __graft_entry__.py
+
+test:
+	python -m pytest tests -q -m "not gpu"
+
+test-gpu:
+	python -m pytest tests -q -m gpu
+
+bench:
+	python bench.py --gpus 1 --steps 20 --warmup 5
+
+lint:
+	python -m pycodestyle --max-line-length=160 tensorflowonspark_amd || true
+
+clean:
+	rm -rf build tensorflowonspark_amd/ops/tfosr_hip_ops.so tools/bin
