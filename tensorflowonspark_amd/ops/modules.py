@@ -360,7 +360,9 @@ class Conv1x1(nn.Module):
     def forward(self, x):
         import os
         backend = os.environ.get("TFOS_CONV1X1", "mfma")
-        if x.is_cuda and backend != "miopen":
+        # the MFMA kernel is bf16-only: a silent downcast would corrupt a
+        # claimed-fp32 run, so other dtypes use the library conv
+        if x.is_cuda and x.dtype == torch.bfloat16 and backend != "miopen":
             x = x.contiguous(memory_format=torch.channels_last)
             if backend == "blas":
                 N, Cin, H, W = x.shape
