@@ -1,0 +1,93 @@
+#!/usr/bin/env python3
+"""MNIST via InputMode.TENSORFLOW — workers read TFRecord shards directly
+(parity: reference ``examples/mnist/keras/mnist_tf_ds.py``, which read
+TFRecords from HDFS via ``ctx.absolute_path``). Sync DDP across workers; no
+feeding job.
+
+  python examples/mnist/mnist_data_setup.py --output data/mnist --format tfr
+  python examples/mnist/mnist_tf_ds.py --cluster_size 2 --num_gpus 0
+"""
+
+import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", ".."))
+
+
+def map_fun(args, ctx):
+    import numpy as np
+    import torch
+
+    from tensorflowonspark_amd.models import MNISTNet
+    from tensorflowonspark_amd.ops.modules import BucketSGD, softmax_cross_entropy
+    from tensorflowonspark_amd.parallel import DDPEngine
+    from tensorflowonspark_amd.utils import checkpoint as ckpt
+    from tensorflowonspark_amd.utils.dataset import TFRecordDataset
+
+    ctx.init_process_group()
+    device = ctx.device
+    model = MNISTNet().to(device)
+    engine = DDPEngine(model, bucket_mb=4)
+    opt = BucketSGD(engine, lr=args.lr, momentum=0.9)
+    model.train()
+
+    data_dir = ctx.absolute_path(args.data_dir)
+    ds = TFRecordDataset(data_dir, ctx.task_index, ctx.world_size,
+                         batch_size=args.batch_size, shuffle_buffer=512)
+    for _epoch in range(args.epochs):
+        it = iter(ds)
+        while True:
+            batch = next(it, None)
+            if not engine.all_ranks_ready(batch is not None):
+                break
+            img = np.asarray([ex["image"][1] for ex in batch],
+                             dtype=np.float32) / 255.0
+            lab = np.asarray([ex["label"][1][0] for ex in batch],
+                             dtype=np.int64)
+            x = torch.as_tensor(img, device=device).reshape(-1, 1, 28, 28)
+            y = torch.as_tensor(lab, device=device)
+            opt.zero_grad()
+            loss = softmax_cross_entropy(model(x), y)
+            loss.backward()
+            engine.finalize_backward()
+            opt.step()
+    if ctx.is_chief:
+        ckpt.save_checkpoint(args.model_dir, args.epochs, model)
+        ctx.export_saved_model(model.cpu(), args.export_dir)
+    torch.distributed.destroy_process_group()
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--cluster_size", type=int, default=2)
+    p.add_argument("--epochs", type=int, default=1)
+    p.add_argument("--batch_size", type=int, default=64)
+    p.add_argument("--lr", type=float, default=0.01)
+    p.add_argument("--data_dir", default="data/mnist/tfr")
+    p.add_argument("--model_dir", default="mnist_tf_model")
+    p.add_argument("--export_dir", default="mnist_tf_export")
+    p.add_argument("--num_gpus", type=int, default=1)
+    args = p.parse_args()
+
+    if not os.path.isdir(args.data_dir):
+        os.system("{} {} --output {} --format tfr --num 2000".format(
+            sys.executable,
+            os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                         "mnist_data_setup.py"),
+            os.path.dirname(args.data_dir)))
+    args.data_dir = os.path.abspath(args.data_dir)
+
+    from tensorflowonspark_amd import TFCluster
+    from tensorflowonspark_amd.local_context import LocalSparkContext
+    sc = LocalSparkContext(num_executors=args.cluster_size)
+    cluster = TFCluster.run(sc, map_fun, args, args.cluster_size,
+                            input_mode=TFCluster.InputMode.TENSORFLOW,
+                            master_node="chief", num_gpus=args.num_gpus)
+    cluster.shutdown(grace_secs=2)
+    sc.stop()
+    print("done; export:", args.export_dir)
+
+
+if __name__ == "__main__":
+    main()
