@@ -188,6 +188,7 @@ class TFSparkNode(object):
     hazard when bootstrap and feeder tasks share one executor process."""
     mgr = None
     owned_mgr = None
+    owned_ring = None
     cluster_id = None
 
 
@@ -346,6 +347,7 @@ def run(fn, tf_args, cluster_meta, tensorboard=False, log_dir=None, queues=None,
                 mgr.set("ring_name", ring_name)
                 mgr.set("ring_slots", slots)
                 mgr.set("ring_slot_bytes", slot_bytes)
+                TFSparkNode.owned_ring = ring
             except Exception as e:
                 logger.warning("shared-memory ring unavailable (%s); "
                                "falling back to inline row blocks", e)

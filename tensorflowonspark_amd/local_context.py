@@ -58,6 +58,9 @@ def _executor_main(exec_id, workdir, task_q, result_q, env):
             # process outlives the context (it would hold stdio pipes open)
             try:
                 from .TFSparkNode import TFSparkNode
+                if TFSparkNode.owned_ring is not None:
+                    TFSparkNode.owned_ring.close()
+                    TFSparkNode.owned_ring.unlink()
                 if TFSparkNode.owned_mgr is not None:
                     TFSparkNode.owned_mgr.shutdown()
             except Exception:
