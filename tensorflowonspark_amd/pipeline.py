@@ -101,13 +101,15 @@ HasExportDir = _mixin("exportDir", "export_dir")
 HasSignatureDefKey = _mixin("signatureDefKey", "signature_def_key")
 HasTagSet = _mixin("tagSet", "tag_set")
 HasNumGPUs = _mixin("numGPUs", "num_gpus", 1)
+HasInputShapes = _mixin("inputShapes", "input_shapes")
 
 
 class TFParams(_Params, HasBatchSize, HasClusterSize, HasEpochs, HasGraceSecs,
                HasInputMapping, HasInputMode, HasMasterNode, HasModelDir,
                HasNumPS, HasDriverPSNodes, HasOutputMapping, HasProtocol,
                HasReaders, HasSteps, HasTensorboard, HasTFRecordDir,
-               HasExportDir, HasSignatureDefKey, HasTagSet, HasNumGPUs):
+               HasExportDir, HasSignatureDefKey, HasTagSet, HasNumGPUs,
+               HasInputShapes):
     """Merged args+params holder (parity: reference ``pipeline.py:342-351``)."""
 
     def __init__(self, tf_args=None):
@@ -236,10 +238,18 @@ def _run_model(iterator, args):
     input_cols = sorted(args.input_mapping) if args.input_mapping else None
     n_inputs = len(input_cols) if input_cols else 1
     batch_size = args.batch_size or 100
+    shapes = args.input_shapes or {}
     for tensors in yield_batch(iterator, batch_size, n_inputs):
         inputs = []
-        for col_vals in tensors:
+        for i, col_vals in enumerate(tensors):
             t = torch.as_tensor(col_vals, dtype=torch.float32, device=device)
+            # coerce flat row arrays to the expected tensor shape (the
+            # reference reshaped from the saved_model signature,
+            # pipeline.py:618-630; TorchScript has no shape metadata, so the
+            # shape comes from the input_shapes param)
+            col = input_cols[i] if input_cols else None
+            if col in shapes:
+                t = t.reshape([t.shape[0]] + list(shapes[col]))
             inputs.append(t)
         with torch.no_grad():
             out = model(*inputs)

@@ -88,3 +88,61 @@ def test_estimator_fit_transform(tmp_path):
             assert p[0] == pytest.approx(e, abs=0.05), (p, e)
     finally:
         sc.stop()
+
+
+def _cnn_train_fn(args, ctx):
+    import torch
+
+    from tensorflowonspark_amd.models import MNISTNet
+    from tensorflowonspark_amd.ops.modules import BucketSGD, softmax_cross_entropy
+    from tensorflowonspark_amd.parallel import DDPEngine
+    ctx.init_process_group(backend="gloo")
+    torch.manual_seed(0)
+    model = MNISTNet()
+    engine = DDPEngine(model, bucket_mb=2)
+    opt = BucketSGD(engine, lr=0.05, momentum=0.9)
+    feed = ctx.get_data_feed(train_mode=True)
+    while True:
+        batch = feed.next_batch(32) if not feed.should_stop() else []
+        if not engine.all_ranks_ready(len(batch) > 0):
+            break
+        x = torch.tensor([r[0] for r in batch]).float().reshape(-1, 1, 28, 28) / 255
+        y = torch.tensor([r[1] for r in batch])
+        opt.zero_grad()
+        loss = softmax_cross_entropy(model(x), y)
+        loss.backward()
+        engine.finalize_backward()
+        opt.step()
+    feed.terminate()
+    if ctx.is_chief:
+        ctx.export_saved_model(model.cpu(), args.export_dir)
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_transform_with_input_shapes(tmp_path):
+    """Flat image arrays reshaped via the input_shapes param during transform
+    (reference coerced shapes from the saved_model signature)."""
+    import numpy as np
+    sc = LocalSparkContext(num_executors=2)
+    try:
+        rng = np.random.default_rng(0)
+        rows = [(rng.integers(0, 256, 784).tolist(), int(rng.integers(0, 10)))
+                for _ in range(200)]
+        df = sc.createDataFrame(rows, ["image", "label"])
+        export_dir = str(tmp_path / "export")
+        est = TFEstimator(_cnn_train_fn, {"export_dir": export_dir}) \
+            .setClusterSize(2).setEpochs(1).setBatchSize(32) \
+            .setInputMapping({"image": "x", "label": "y"})
+        model = est.fit(df)
+
+        tdf = sc.createDataFrame([(r[0],) for r in rows[:8]], ["image"])
+        model.setInputMapping({"image": "x"}) \
+             .setOutputMapping({"logits": "prediction"}) \
+             .setInputShapes({"image": [1, 28, 28]})
+        preds = model.transform(tdf).collect()
+        assert len(preds) == 8
+        assert len(preds[0][0]) == 10  # 10 logits per row
+    finally:
+        sc.stop()
