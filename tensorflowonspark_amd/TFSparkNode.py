@@ -497,9 +497,9 @@ def run(fn, tf_args, cluster_meta, tensorboard=False, log_dir=None, queues=None,
     return _mapfn
 
 
-def _feed_partition(mgr, iterator, block_rows):
+def _feed_partition(mgr, iterator, block_rows, qname="input"):
     """Pack an RDD partition into ring blocks (or inline blocks w/o shm)."""
-    queue = mgr.get_queue("input")
+    queue = mgr.get_queue(qname)
     ring = None
     ring_name = mgr.get("ring_name")
     if ring_name is not None:
@@ -558,7 +558,7 @@ def train(cluster_info, cluster_meta, feed_timeout=600, qname="input"):
             logger.info("terminating: skipped %d rows", count)
         else:
             block_rows = int(cluster_meta.get("block_rows", 512)) if cluster_meta else 512
-            count = _feed_partition(mgr, iterator, block_rows)
+            count = _feed_partition(mgr, iterator, block_rows, qname)
             logger.info("fed %d rows", count)
             # wait for the consumer to finish this partition, polling errors
             joined = [False]
@@ -606,7 +606,7 @@ def inference(cluster_info, feed_timeout=600, qname="input"):
         mgr = _get_manager(cluster_info, host, executor_id)
         queue_in = mgr.get_queue(qname)
 
-        count = _feed_partition(mgr, iterator, 512)
+        count = _feed_partition(mgr, iterator, 512, qname)
         queue_in.put(("end_partition",))
         if count == 0:
             # consume our own marker so join() can complete

@@ -144,3 +144,27 @@ def test_uneven_feed_through_cluster(sc):
         time.sleep(0.5)
     assert len(steps) == 2
     assert steps[0] == steps[1], steps  # ranks stopped together
+
+
+def _multi_queue_fn(args, ctx):
+    """Consume from a custom queue name (reference supports arbitrary queues
+    via TFCluster.run(queues=[...]) + qname on train/inference)."""
+    feed = ctx.get_data_feed(train_mode=False, qname_in="alt_in",
+                             qname_out="alt_out")
+    while not feed.should_stop():
+        batch = feed.next_batch(10)
+        if not batch:
+            break
+        feed.batch_results([x + 1000 for x in batch])
+
+
+def test_custom_queue_names(sc):
+    cluster = TFCluster.run(sc, _multi_queue_fn, {}, num_executors=2, num_ps=0,
+                            master_node=None,
+                            input_mode=TFCluster.InputMode.SPARK, num_gpus=0,
+                            queues=["alt_in", "alt_out", "error"],
+                            reservation_timeout=60)
+    rdd = sc.parallelize(range(100), 2)
+    out = cluster.inference(rdd, qname="alt_in").collect()
+    assert sorted(out) == [x + 1000 for x in range(100)]
+    cluster.shutdown(grace_secs=0)
