@@ -72,14 +72,16 @@ class TFCluster(object):
             TFSparkNode.train(self.cluster_info, self.cluster_meta,
                               feed_timeout=feed_timeout, qname=qname))
 
-    def inference(self, dataRDD, feed_timeout=600, qname="input"):
+    def inference(self, dataRDD, feed_timeout=600, qname="input",
+                  qname_out="output"):
         """Feed an RDD for inference; returns an RDD of results (1:1 with rows)."""
         logger.info("starting inference")
         assert self.input_mode == InputMode.SPARK, "inference() requires InputMode.SPARK"
         assert qname in self.queues, "unknown queue: {}".format(qname)
+        assert qname_out in self.queues, "unknown queue: {}".format(qname_out)
         return dataRDD.mapPartitions(
             TFSparkNode.inference(self.cluster_info, feed_timeout=feed_timeout,
-                                  qname=qname))
+                                  qname=qname, qname_out=qname_out))
 
     def shutdown(self, ssc=None, grace_secs=0, timeout=259200):
         """Stop the cluster; raise any error trapped in worker error queues.

@@ -595,7 +595,7 @@ def train(cluster_info, cluster_meta, feed_timeout=600, qname="input"):
     return _train
 
 
-def inference(cluster_info, feed_timeout=600, qname="input"):
+def inference(cluster_info, feed_timeout=600, qname="input", qname_out="output"):
     """Factory: mapPartitions closure producing exactly one result per input row."""
 
     def _inference(iterator):
@@ -636,7 +636,7 @@ def inference(cluster_info, feed_timeout=600, qname="input"):
         if count == 0:
             return []
         # pop exactly `count` results (reference invariant TFSparkNode.py:587-594)
-        queue_out = mgr.get_queue("output")
+        queue_out = mgr.get_queue(qname_out)
         results = []
         while len(results) < count:
             results.append(queue_out.get(block=True))

@@ -165,6 +165,6 @@ def test_custom_queue_names(sc):
                             queues=["alt_in", "alt_out", "error"],
                             reservation_timeout=60)
     rdd = sc.parallelize(range(100), 2)
-    out = cluster.inference(rdd, qname="alt_in").collect()
+    out = cluster.inference(rdd, qname="alt_in", qname_out="alt_out").collect()
     assert sorted(out) == [x + 1000 for x in range(100)]
     cluster.shutdown(grace_secs=0)
