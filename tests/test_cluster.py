@@ -168,3 +168,36 @@ def test_custom_queue_names(sc):
     out = cluster.inference(rdd, qname="alt_in", qname_out="alt_out").collect()
     assert sorted(out) == [x + 1000 for x in range(100)]
     cluster.shutdown(grace_secs=0)
+
+
+def _port_check_fn(args, ctx):
+    """release_port=False: the reserved port stays bound until the user frees
+    it (reference tests/test_TFCluster.py:93-121 semantics)."""
+    import socket
+    held = ctx.tmp_socket is not None
+    bound_before = False
+    if held:
+        try:
+            probe = socket.socket()
+            probe.bind(("127.0.0.1", ctx._port))
+            probe.close()
+        except OSError:
+            bound_before = True   # port is occupied by our reservation
+    ctx.release_port()
+    released = ctx.tmp_socket is None
+    feed = ctx.get_data_feed(train_mode=False)
+    while not feed.should_stop():
+        batch = feed.next_batch(10)
+        if not batch:
+            break
+        feed.batch_results([(held, bound_before, released)] * len(batch))
+
+
+def test_release_port_deferred(sc):
+    cluster = TFCluster.run(sc, _port_check_fn, {}, num_executors=2, num_ps=0,
+                            master_node=None, release_port=False,
+                            input_mode=TFCluster.InputMode.SPARK,
+                            num_gpus=0, reservation_timeout=60)
+    out = cluster.inference(sc.parallelize(range(4), 2)).collect()
+    assert all(held and bound and released for held, bound, released in out), out
+    cluster.shutdown(grace_secs=0)
