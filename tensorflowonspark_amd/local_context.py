@@ -289,7 +289,11 @@ class LocalSparkContext:
         self._result_q = ctx.Queue()
         self._task_qs = []
         self._procs = []
-        base_env = {"TFOS_FORCE_LOOPBACK": "1", "TFOS_FORCE_LOOPBACK_MASTER": "1"}
+        # cap math-library threads: N executor processes each defaulting to
+        # all cores oversubscribes the host badly during CPU runs
+        base_env = {"TFOS_FORCE_LOOPBACK": "1", "TFOS_FORCE_LOOPBACK_MASTER": "1",
+                    "OMP_NUM_THREADS": os.environ.get("OMP_NUM_THREADS", "4"),
+                    "MKL_NUM_THREADS": os.environ.get("MKL_NUM_THREADS", "4")}
         base_env.update(env or {})
         for i in range(num_executors):
             tq = ctx.Queue()

@@ -78,7 +78,7 @@ class TFRecordDataset:
         files = list(self.files)
         if self.shuffle_buffer:
             import random
-            rng = random.Random((self.seed, self._epoch))
+            rng = random.Random(hash((self.seed, self._epoch)))
             rng.shuffle(files)
             buf = []
             for ex in tfrecord_examples(files, self.decode):

@@ -120,7 +120,7 @@ def _cnn_train_fn(args, ctx):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(600)
 def test_transform_with_input_shapes(tmp_path):
     """Flat image arrays reshaped via the input_shapes param during transform
     (reference coerced shapes from the saved_model signature)."""
@@ -129,7 +129,7 @@ def test_transform_with_input_shapes(tmp_path):
     try:
         rng = np.random.default_rng(0)
         rows = [(rng.integers(0, 256, 784).tolist(), int(rng.integers(0, 10)))
-                for _ in range(200)]
+                for _ in range(96)]
         df = sc.createDataFrame(rows, ["image", "label"])
         export_dir = str(tmp_path / "export")
         est = TFEstimator(_cnn_train_fn, {"export_dir": export_dir}) \
