@@ -285,6 +285,11 @@ class LocalSparkContext:
         self.defaultParallelism = num_executors
         self._root = workdir_root or tempfile.mkdtemp(prefix="tfos_local_")
         self._own_root = workdir_root is None
+        # fork context: executors inherit imports cheaply and closures don't
+        # need spawn-compatible __main__. Create the context BEFORE running
+        # torch compute in the driver — forking after OpenMP/ATen thread
+        # pools exist is the classic libgomp post-fork deadlock. (Real Spark
+        # executors are separate JVM-launched processes and are unaffected.)
         ctx = multiprocessing.get_context("fork")
         self._result_q = ctx.Queue()
         self._task_qs = []
