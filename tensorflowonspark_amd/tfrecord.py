@@ -150,9 +150,9 @@ def _len_delim(field, payload):
     return _tag(field, 2) + _varint(len(payload)) + payload
 
 
-def _zigzag_int64(n):
+def _uint64(n):
     # int64 values are stored as plain varints (two's complement, 10 bytes if
-    # negative) in Int64List — not zigzag
+    # negative) in Int64List — protobuf int64, not sint64/zigzag
     return n & 0xFFFFFFFFFFFFFFFF
 
 
@@ -172,7 +172,7 @@ def _feature_floats(values):
 
 
 def _feature_int64s(values):
-    packed = b"".join(_varint(_zigzag_int64(int(v))) for v in values)
+    packed = b"".join(_varint(_uint64(int(v))) for v in values)
     payload = _len_delim(1, packed)           # Int64List.value packed
     return _len_delim(3, payload)             # Feature.int64_list = 3
 
