@@ -286,7 +286,13 @@ at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
 
 }  // namespace
 
+// bump when the binding surface changes: the Python side refuses to run
+// against a stale in-tree .so (clear "rebuild" message instead of a random
+// AttributeError mid-training)
+#define TFOSR_API_VERSION 3
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("api_version", []() { return TFOSR_API_VERSION; });
   m.def("bn_fwd_train", &bn_fwd_train);
   m.def("bn_fwd_eval", &bn_fwd_eval);
   m.def("bn_bwd", &bn_bwd);

@@ -17,6 +17,9 @@ import torch
 
 logger = logging.getLogger(__name__)
 
+#: must match TFOSR_API_VERSION in csrc/bindings.cpp
+API_VERSION = 3
+
 _ext = None
 _ext_checked = False
 
@@ -46,6 +49,12 @@ def get_ext(required=False):
                 spec = importlib.util.spec_from_file_location("tfosr_hip_ops", path)
                 mod = importlib.util.module_from_spec(spec)
                 spec.loader.exec_module(mod)
+                got = mod.api_version() if hasattr(mod, "api_version") else 0
+                if got != API_VERSION:
+                    raise RuntimeError(
+                        "tfosr_hip_ops.so is stale (api {} != {}): rebuild "
+                        "with `python __graft_entry__.py`".format(
+                            got, API_VERSION))
                 _ext = mod
                 logger.info("loaded HIP ops extension: %s", path)
             except Exception as e:
