@@ -149,3 +149,62 @@ def test_uneven_partition_guard():
         assert p.exitcode == 0
     # both ranks stopped together after the shorter feed (3 steps)
     assert results == {0: 3, 1: 3}
+
+
+def test_no_sync_gradient_accumulation():
+    """no_sync() skips bucket reduction; grads accumulate across micro-steps
+    and one synced step matches a single large-batch step."""
+    torch.manual_seed(3)
+    data = torch.randn(32, 784)
+    target = torch.randint(0, 10, (32,))
+
+    m1 = MNISTMLP(hidden=16)
+    m2 = MNISTMLP(hidden=16)
+    m2.load_state_dict(m1.state_dict())
+
+    # big-batch reference
+    e1 = DDPEngine(m1, bucket_mb=1)
+    o1 = BucketSGD(e1, lr=0.1, momentum=0.0)
+    o1.zero_grad()
+    loss = softmax_cross_entropy(m1(data), target, reduction="sum") / 32
+    loss.backward()
+    e1.finalize_backward()
+    o1.step()
+
+    # two accumulated micro-batches
+    e2 = DDPEngine(m2, bucket_mb=1)
+    o2 = BucketSGD(e2, lr=0.1, momentum=0.0)
+    o2.zero_grad()
+    with e2.no_sync():
+        l1 = softmax_cross_entropy(m2(data[:16]), target[:16], reduction="sum") / 32
+        l1.backward()
+    l2 = softmax_cross_entropy(m2(data[16:]), target[16:], reduction="sum") / 32
+    l2.backward()
+    e2.finalize_backward()
+    o2.step()
+
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-6)
+
+
+def test_bucket_sgd_nesterov():
+    torch.manual_seed(4)
+    m1 = MNISTMLP(hidden=8)
+    m2 = MNISTMLP(hidden=8)
+    m2.load_state_dict(m1.state_dict())
+    ref = torch.optim.SGD(m1.parameters(), lr=0.05, momentum=0.9, nesterov=True)
+    e = DDPEngine(m2, bucket_mb=1)
+    opt = BucketSGD(e, lr=0.05, momentum=0.9, nesterov=True)
+    x = torch.randn(8, 784)
+    y = torch.randint(0, 10, (8,))
+    for _ in range(3):
+        ref.zero_grad()
+        torch.nn.functional.cross_entropy(m1(x), y).backward()
+        ref.step()
+        opt.zero_grad()
+        loss = softmax_cross_entropy(m2(x), y)
+        loss.backward()
+        e.finalize_backward()
+        opt.step()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-5), (p1 - p2).abs().max()

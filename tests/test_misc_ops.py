@@ -78,3 +78,28 @@ def test_piecewise_and_cosine_schedules():
     assert s(11) == pytest.approx(0.2)
     c = CosineLR(1.0, 10)
     assert c(5) == pytest.approx(0.5, abs=1e-6)
+
+
+def test_bucket_adamw_matches_torch():
+    from tensorflowonspark_amd.ops.modules import BucketAdam
+    from tensorflowonspark_amd.parallel import DDPEngine
+    torch.manual_seed(5)
+    m1 = torch.nn.Linear(6, 3)
+    m2 = torch.nn.Linear(6, 3)
+    m2.load_state_dict(m1.state_dict())
+    ref = torch.optim.AdamW(m1.parameters(), lr=0.01, weight_decay=0.05)
+    e = DDPEngine(m2, bucket_mb=1)
+    opt = BucketAdam(e, lr=0.01, weight_decay=0.05, decoupled=True)
+    x = torch.randn(12, 6)
+    y = torch.randn(12, 3)
+    for _ in range(4):
+        ref.zero_grad()
+        torch.nn.functional.mse_loss(m1(x), y).backward()
+        ref.step()
+        opt.zero_grad()
+        loss = torch.nn.functional.mse_loss(m2(x), y)
+        loss.backward()
+        e.finalize_backward()
+        opt.step()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-6), (p1 - p2).abs().max()
