@@ -102,7 +102,13 @@ def fromTFExample(record, binary_features=None, schema=None):
 
 
 def saveAsTFRecords(df, output_dir):
-    """Save a DataFrame as TFRecord part files under ``output_dir``."""
+    """Save a DataFrame as TFRecord part files under ``output_dir``.
+
+    Part files are written executor-side via the native codec, so
+    ``output_dir`` must be a path every executor can reach (shared/posix
+    filesystem or fuse-mounted object store). The reference reached HDFS
+    through the tensorflow-hadoop OutputFormat (``dfutil.py:39-41``); direct
+    hdfs:// output without a mount is not supported here."""
     dtypes = df.dtypes
     convert = toTFExample(dtypes)
     os.makedirs(output_dir, exist_ok=True)
