@@ -46,6 +46,11 @@ def main():
     p.add_argument("--data", default="data/mnist/mnist.csv")
     p.add_argument("--export_dir", default="mnist_export")
     args = p.parse_args()
+    # executor working dirs differ from the driver's: path args
+    # must be absolute (shared-filesystem semantics, as on a real
+    # cluster)
+    args.data = os.path.abspath(args.data)
+    args.export_dir = os.path.abspath(args.export_dir)
 
     from tensorflowonspark_amd import TFParallel
     from tensorflowonspark_amd.local_context import LocalSparkContext

@@ -48,6 +48,10 @@ def main():
     p.add_argument("--export_dir", default="mnist_pipeline_export")
     p.add_argument("--num", type=int, default=2000)
     args = p.parse_args()
+    # executor working dirs differ from the driver's: path args
+    # must be absolute (shared-filesystem semantics, as on a real
+    # cluster)
+    args.export_dir = os.path.abspath(args.export_dir)
 
     import importlib.util
     spec = importlib.util.spec_from_file_location(

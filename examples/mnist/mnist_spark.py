@@ -81,6 +81,11 @@ def main():
     p.add_argument("--num_gpus", type=int, default=1)
     p.add_argument("--use_pyspark", action="store_true")
     args = p.parse_args()
+    # executor working dirs differ from the driver's: path args
+    # must be absolute (shared-filesystem semantics, as on a real
+    # cluster)
+    args.model_dir = os.path.abspath(args.model_dir)
+    args.export_dir = os.path.abspath(args.export_dir)
 
     from tensorflowonspark_amd import TFCluster
     if args.use_pyspark:
@@ -92,10 +97,15 @@ def main():
 
     if not os.path.exists(args.data):
         print("generating data (run mnist_data_setup.py for more control)")
+        out_dir = os.path.dirname(args.data) or "."
         os.system("{} {} --output {} --format csv".format(
             sys.executable,
-            os.path.join(os.path.dirname(__file__), "mnist_data_setup.py"),
-            os.path.dirname(args.data)))
+            os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                         "mnist_data_setup.py"), out_dir))
+        generated = os.path.join(out_dir, "mnist.csv")
+        if not os.path.exists(args.data) and os.path.exists(generated):
+            args.data = generated  # setup always writes mnist.csv
+    args.data = os.path.abspath(args.data)
 
     rows = []
     with open(args.data) as f:

@@ -69,6 +69,11 @@ def main():
     p.add_argument("--export_dir", default="mnist_tf_export")
     p.add_argument("--num_gpus", type=int, default=1)
     args = p.parse_args()
+    # executor working dirs differ from the driver's: path args
+    # must be absolute (shared-filesystem semantics, as on a real
+    # cluster)
+    args.model_dir = os.path.abspath(args.model_dir)
+    args.export_dir = os.path.abspath(args.export_dir)
 
     if not os.path.isdir(args.data_dir):
         os.system("{} {} --output {} --format tfr --num 2000".format(

@@ -85,6 +85,10 @@ def main():
     p.add_argument("--num_gpus", type=int, default=1)
     p.add_argument("--model_dir", default="resnet_model")
     args = p.parse_args()
+    # executor working dirs differ from the driver's: path args
+    # must be absolute (shared-filesystem semantics, as on a real
+    # cluster)
+    args.model_dir = os.path.abspath(args.model_dir)
 
     import numpy as np
 

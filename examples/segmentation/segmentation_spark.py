@@ -69,6 +69,10 @@ def main():
     p.add_argument("--num_gpus", type=int, default=1)
     p.add_argument("--export_dir", default="segmentation_export")
     args = p.parse_args()
+    # executor working dirs differ from the driver's: path args
+    # must be absolute (shared-filesystem semantics, as on a real
+    # cluster)
+    args.export_dir = os.path.abspath(args.export_dir)
 
     import numpy as np
 
