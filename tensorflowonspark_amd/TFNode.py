@@ -50,12 +50,16 @@ def start_cluster_server(ctx, num_gpus=1, rdma=False):
         "— collectives run over RCCL/xGMI via torch.distributed.")
 
 
-def export_saved_model(model, export_dir, is_chief=True):
+def export_saved_model(model, export_dir, is_chief=True, require_script=False):
     """Chief-only model export (layout parity: versioned export_dir).
 
-    Saves a TorchScript trace when possible, else the state_dict, under
+    Saves a TorchScript export when possible, else the state_dict, under
     ``export_dir``. Non-chief ranks write nothing (reference ``compat.py:10-17``
     had non-chief write to a dummy path; skipping is the cleaner equivalent).
+
+    A state_dict fallback cannot be consumed by ``TFModel.transform`` (it
+    needs TorchScript); pass ``require_script=True`` to fail *here* instead of
+    at transform time.
     """
     if not is_chief:
         return None
@@ -65,9 +69,17 @@ def export_saved_model(model, export_dir, is_chief=True):
         scripted = torch.jit.script(model)
         path = os.path.join(export_dir, "model.pt")
         scripted.save(path)
-    except Exception:
+    except Exception as e:
+        if require_script:
+            raise RuntimeError(
+                "torch.jit.script failed for this model and the export is "
+                "required to be TorchScript (TFModel.transform consumes it): "
+                "{}".format(e)) from e
         path = os.path.join(export_dir, "state_dict.pt")
         torch.save(model.state_dict(), path)
+        logger.warning(
+            "torch.jit.script failed (%s); exported state_dict only to %s — "
+            "TFModel.transform will NOT be able to load this export", e, path)
     logger.info("exported model to %s", path)
     return path
 
@@ -96,7 +108,11 @@ class DataFeed(object):
         self.qname_in = qname_in
         self.qname_out = qname_out
         self.done_feeding = False
-        self.input_tensors = sorted(input_mapping.values()) if input_mapping else None
+        # Row values arrive in column-sorted order (the feeder selects columns
+        # sorted by name, pipeline.py:147); bind tensor names in that same
+        # order, not sorted by tensor name (reference TFNode.py:251).
+        self.input_tensors = [t for _c, t in sorted(input_mapping.items())] \
+            if input_mapping else None
         self._queue_in = mgr.get_queue(qname_in)
         self._queue_out = mgr.get_queue(qname_out)
         self._buffer = deque()

@@ -104,17 +104,19 @@ def fromTFExample(record, binary_features=None, schema=None):
 def saveAsTFRecords(df, output_dir):
     """Save a DataFrame as TFRecord part files under ``output_dir``.
 
-    Part files are written executor-side via the native codec, so
-    ``output_dir`` must be a path every executor can reach (shared/posix
-    filesystem or fuse-mounted object store). The reference reached HDFS
-    through the tensorflow-hadoop OutputFormat (``dfutil.py:39-41``); direct
-    hdfs:// output without a mount is not supported here."""
+    Part files are written executor-side via the native codec. ``output_dir``
+    may be a plain/shared posix path or a schemed URI (hdfs://, file://,
+    s3a://, ...) opened through fsio/fsspec — the capability the reference got
+    from the tensorflow-hadoop OutputFormat (``dfutil.py:39-41``)."""
+    from .utils import fsio
     dtypes = df.dtypes
     convert = toTFExample(dtypes)
-    os.makedirs(output_dir, exist_ok=True)
+    fsio.fs_makedirs(output_dir)
 
     def _write(idx, iterator):
-        path = os.path.join(output_dir, "part-r-{:05d}".format(idx))
+        path = output_dir.rstrip("/") + "/part-r-{:05d}".format(idx) \
+            if fsio.get_scheme(output_dir) else \
+            os.path.join(output_dir, "part-r-{:05d}".format(idx))
         n = 0
         with tfrecord.TFRecordWriter(path) as w:
             for rec in convert(iterator):
@@ -127,19 +129,27 @@ def saveAsTFRecords(df, output_dir):
         counts = rdd.mapPartitionsWithIndex(_write).collect()
         logger.info("wrote %s records to %s", sum(counts), output_dir)
     else:  # pragma: no cover
-        with tfrecord.TFRecordWriter(os.path.join(output_dir, "part-r-00000")) as w:
+        single = output_dir.rstrip("/") + "/part-r-00000" \
+            if fsio.get_scheme(output_dir) else \
+            os.path.join(output_dir, "part-r-00000")
+        with tfrecord.TFRecordWriter(single) as w:
             for rec in convert(iter(df.collect())):
                 w.write(rec)
 
 
 def loadTFRecords(sc, input_dir, binary_features=None):
-    """Load TFRecord files under ``input_dir`` as a DataFrame with inferred
-    schema; records provenance in ``loadedDF``."""
-    files = sorted(f for f in glob.glob(os.path.join(input_dir, "part-*"))
-                   if os.path.isfile(f))
-    if not files:
-        files = sorted(f for f in glob.glob(os.path.join(input_dir, "*"))
-                       if os.path.isfile(f) and not os.path.basename(f).startswith("_"))
+    """Load TFRecord files under ``input_dir`` (plain path or schemed URI)
+    as a DataFrame with inferred schema; records provenance in ``loadedDF``."""
+    from .utils import fsio
+    if fsio.get_scheme(input_dir):
+        files = fsio.fs_glob(input_dir.rstrip("/") + "/part-*") or \
+            fsio.fs_listfiles(input_dir)
+    else:
+        files = sorted(f for f in glob.glob(os.path.join(input_dir, "part-*"))
+                       if os.path.isfile(f))
+        if not files:
+            files = sorted(f for f in glob.glob(os.path.join(input_dir, "*"))
+                           if os.path.isfile(f) and not os.path.basename(f).startswith("_"))
     if not files:
         raise FileNotFoundError("no TFRecord files under " + input_dir)
 
@@ -151,16 +161,29 @@ def loadTFRecords(sc, input_dir, binary_features=None):
             for rec in tfrecord.tfrecord_iterator(path):
                 yield fromTFExample(rec, binary_features, schema)
 
-    rows = sc.parallelize(files, min(len(files), sc.defaultParallelism)) \
-        .mapPartitions(_read).collect()
+    rdd = sc.parallelize(files, min(len(files), sc.defaultParallelism)) \
+        .mapPartitions(_read)
     if hasattr(sc, "createDataFrame"):
-        df = sc.createDataFrame(rows, [n for n, _ in schema],
+        # LocalSparkContext path: small data, driver-side assembly
+        df = sc.createDataFrame(rdd.collect(), [n for n, _ in schema],
                                 [t for _, t in schema])
     else:  # pragma: no cover - real pyspark path
+        # keep the load distributed (no driver-side collect — the reference's
+        # TFRecordFileInputFormat load was distributed too, dfutil.py:63-65)
         from pyspark.sql import SparkSession
+        from pyspark.sql.types import (ArrayType, DoubleType, LongType,
+                                       BinaryType, StringType, StructField,
+                                       StructType)
+        base = {"bigint": LongType(), "double": DoubleType(),
+                "string": StringType(), "binary": BinaryType()}
+        fields = []
+        for n, t in schema:
+            if t.startswith("array<"):
+                fields.append(StructField(n, ArrayType(base[t[6:-1]])))
+            else:
+                fields.append(StructField(n, base[t]))
         spark = SparkSession.builder.getOrCreate()
-        df = spark.createDataFrame(rows, " ".join(
-            "{}:{}".format(n, t) for n, t in schema))
+        df = spark.createDataFrame(rdd, StructType(fields))
     loadedDF[id(df)] = input_dir
     return df
 

@@ -154,6 +154,20 @@ class TFEstimator(TFParams):
                                 num_gpus=args.num_gpus or 0)
         cluster.train(dataset.select(input_cols).rdd, args.epochs)
         cluster.shutdown(grace_secs=args.grace_secs or 30)
+        # Early-fail the export->transform contract: if this pipeline's model
+        # will be loaded by TFModel.transform (export_dir set), the chief must
+        # have produced a TorchScript export — a state_dict-only fallback
+        # would otherwise break the pipeline late, inside transform.
+        if getattr(args, "export_dir", None):
+            script_path = os.path.join(args.export_dir, "model.pt")
+            sd_path = os.path.join(args.export_dir, "state_dict.pt")
+            if not os.path.exists(script_path) and os.path.exists(sd_path):
+                raise RuntimeError(
+                    "fit() produced only a state_dict export at {} — "
+                    "torch.jit.script failed for this model, and "
+                    "TFModel.transform requires TorchScript. Make the model "
+                    "scriptable or export with require_script=True to see "
+                    "the scripting error at export time.".format(sd_path))
         return TFModel(self.merge_args_params())
 
 

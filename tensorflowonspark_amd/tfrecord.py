@@ -56,8 +56,8 @@ def masked_crc(data):
 
 class TFRecordWriter:
     def __init__(self, path):
-        os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
-        self._f = open(path, "wb")
+        from .utils import fsio
+        self._f = fsio.fs_open(path, "wb")  # scheme-aware (hdfs://, file://, ...)
 
     def write(self, record):
         length = struct.pack("<Q", len(record))
@@ -89,11 +89,15 @@ def tfrecord_iterator(path, verify=False):
 
     Uses the C++ codec (HW CRC32-C, single mmap-style scan) when the extension
     is built; the pure-Python path below is the reference implementation."""
+    from .utils import fsio
+    scheme = fsio.get_scheme(path)
+    local = path[len("file://"):] if scheme == "file" else path
     ext = _native_ext()
-    if ext is not None and hasattr(ext, "tfrecord_read_file"):
-        yield from ext.tfrecord_read_file(path, verify)
+    if (scheme is None or scheme == "file") and ext is not None \
+            and hasattr(ext, "tfrecord_read_file"):
+        yield from ext.tfrecord_read_file(local, verify)
         return
-    with open(path, "rb") as f:
+    with fsio.fs_open(path, "rb") as f:
         while True:
             header = f.read(12)
             if len(header) < 12:

@@ -36,7 +36,11 @@ def save_checkpoint(model_dir, step, model, optimizer_state=None, extra=None,
     torch.save(payload, tmp)
     os.replace(tmp, path)  # atomic: a crash never leaves a torn checkpoint
     if keep_last:
-        ckpts = sorted(glob.glob(os.path.join(model_dir, "weights-*.pt")))
+        # sort by numeric step (lexicographic breaks past the 4-digit padding:
+        # 'weights-10000.pt' < 'weights-9999.pt')
+        ckpts = [p for p in glob.glob(os.path.join(model_dir, "weights-*.pt"))
+                 if _PAT.search(p)]
+        ckpts.sort(key=lambda p: int(_PAT.search(p).group(1)))
         for old in ckpts[:-keep_last]:
             try:
                 os.remove(old)

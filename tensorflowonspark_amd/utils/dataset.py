@@ -13,9 +13,20 @@ from .. import tfrecord
 
 
 def shard_files(paths_or_glob, shard_index, num_shards):
-    """Deterministically shard a file list across workers."""
+    """Deterministically shard a file list across workers.
+
+    Accepts a directory, a glob, a schemed URI (hdfs://, file://, ...) or an
+    explicit list; schemed paths are listed through fsio/fsspec."""
+    from . import fsio
     if isinstance(paths_or_glob, str):
-        if os.path.isdir(paths_or_glob):
+        scheme = fsio.get_scheme(paths_or_glob)
+        if scheme is not None:
+            if fsio.fs_isdir(paths_or_glob):
+                files = fsio.fs_glob(paths_or_glob.rstrip("/") + "/part-*") or \
+                    fsio.fs_listfiles(paths_or_glob)
+            else:
+                files = fsio.fs_glob(paths_or_glob)
+        elif os.path.isdir(paths_or_glob):
             files = sorted(glob.glob(os.path.join(paths_or_glob, "part-*"))) or \
                 sorted(f for f in glob.glob(os.path.join(paths_or_glob, "*"))
                        if os.path.isfile(f))
