@@ -42,6 +42,9 @@ void tfosr_maxpool_bwd(const void*, const unsigned char*, void*, int, int, int,
                        int, int, int, int, int, int, int, hipStream_t);
 void tfosr_conv3x3(const void*, const void*, const void*, void*, int, int, int,
                    int, int, int, int, int, int, int, hipStream_t);
+void tfosr_conv_mfma(const void*, const void*, const void*, void*, int, int,
+                     int, int, int, int, int, int, int, int, int, int, int,
+                     hipStream_t);
 void tfosr_conv_wrw(const void*, const void*, float*, int, int, int, int, int,
                     int, int, int, int, int, hipStream_t);
 }
@@ -289,7 +292,7 @@ at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
 // bump when the binding surface changes: the Python side refuses to run
 // against a stale in-tree .so (clear "rebuild" message instead of a random
 // AttributeError mid-training)
-#define TFOSR_API_VERSION 3
+#define TFOSR_API_VERSION 4
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("api_version", []() { return TFOSR_API_VERSION; });
@@ -341,6 +344,36 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     tfosr_conv3x3(x.data_ptr(), w9.contiguous().data_ptr(), guard.data_ptr(),
                   y.data_ptr(), /*out_bf16=*/1, N, H, W, Cin, Cout, OH, OW,
                   S, P, cur_stream());
+    return y;
+  });
+  // Generalized implicit-GEMM conv: taps in {1,9}, stride S, pad P, input
+  // dilation D (the stored input is read as if zero-dilated by D). OH/OW are
+  // explicit because backward-data output size is not derivable from the
+  // stored input dims. One kernel covers conv1x1/conv3x3 fwd at stride 1/2,
+  // their backward-data, and ConvTranspose2d k3 s2 (survey §2.3 rows 1-2).
+  m.def("conv_mfma", [](at::Tensor x, at::Tensor wk, long Cout, long fh,
+                        long fw, long S, long P, long D, long OH, long OW) {
+    TORCH_CHECK(x.dim() == 4 && x.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "conv_mfma expects channels_last input");
+    TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+                wk.scalar_type() == at::kBFloat16);
+    TORCH_CHECK(fh >= 1 && fw >= 1 && fh * fw <= 49, "bad filter dims");
+    TORCH_CHECK(D == 1 || D == 2, "input dilation must be 1 or 2");
+    long taps = fh * fw;
+    int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+    TORCH_CHECK(Cin % 32 == 0, "conv_mfma requires Cin % 32 == 0");
+    TORCH_CHECK(wk.size(1) == taps * Cin);
+    if (OH < 0) {
+      long HV = (H - 1) * D + 1, WV = (W - 1) * D + 1;
+      OH = (HV + 2 * P - fh) / S + 1;
+      OW = (WV + 2 * P - fw) / S + 1;
+    }
+    auto y = at::empty({N, Cout, OH, OW}, x.options(),
+                       at::MemoryFormat::ChannelsLast);
+    auto guard = at::zeros({64}, x.options());
+    tfosr_conv_mfma(x.data_ptr(), wk.contiguous().data_ptr(), guard.data_ptr(),
+                    y.data_ptr(), /*out_bf16=*/1, N, H, W, Cin, Cout, OH, OW,
+                    S, P, taps, fw, D, cur_stream());
     return y;
   });
   m.def("conv_wrw", [](at::Tensor dy, at::Tensor x, long R, long S, long P) {

@@ -13,6 +13,16 @@
 // in the main loop); two tile shapes: 256x256 (Cout >= 192) and 256x128.
 // Backward-data for stride 1 reuses this kernel: dx = conv3x3(dy, W') with
 // W'[cin][r][s][cout] = W[cout][2-r][2-s][cin] (built host-side).
+//
+// Input dilation D (template): the input is read as if zero-dilated by D —
+// virtual index ihv maps to stored row ihv/D and contributes only when
+// ihv % D == 0 (otherwise the tap redirects to the zero guard). This makes
+// the SAME kernel compute:
+//   - stride-2 forward            (S=2, D=1)
+//   - stride-2 backward-data      (S=1, D=2, W' flipped: dx = conv(dy_dil, W'))
+//   - ConvTranspose2d k3 s2       (S=1, D=2, W swapped/flipped)
+// and, with K = 1*Cin (w shaped [Cout, Cin], taps degenerate to r=s=0),
+// 1x1 convolutions at any stride plus their dilated backward-data.
 #include "tfosr_common.h"
 
 typedef short bf16x8 __attribute__((ext_vector_type(8)));
@@ -29,11 +39,12 @@ __device__ __forceinline__ void c3_stage16(const char* src,
 }
 
 // WRG x WCG wave grid (8 waves); per-wave output (MI*16) x 64
-template <typename OT, int WRG, int WCG, int MI>
+template <typename OT, int WRG, int WCG, int MI, int D = 1>
 __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
     const bf16_t* __restrict__ X, const bf16_t* __restrict__ W9,
     const bf16_t* __restrict__ guard, OT* __restrict__ C,
-    int Nn, int H, int Wd, int Cin, int Cout, int OH, int OW, int S, int P) {
+    int Nn, int H, int Wd, int Cin, int Cout, int OH, int OW, int S, int P,
+    int taps, int fw) {
   constexpr int BM = WRG * MI * 16;
   constexpr int BN = WCG * 64;
   constexpr int ABYTES = BM * C3_BK * 2;
@@ -45,7 +56,7 @@ __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
   __attribute__((address_space(3))) char* lds3 =
       (__attribute__((address_space(3))) char*)lds;
 
-  const int K = 9 * Cin;
+  const int K = taps * Cin;
   const long M = (long)Nn * OH * OW;
   const int ntn = (Cout + BN - 1) / BN;
   const int nwg = gridDim.x;
@@ -99,15 +110,21 @@ __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
   auto stage_tile = [&](int slot, int kt) {
     const int k0 = kt * C3_BK;
     const int rs = k0 / Cin;
-    const int rr = rs / 3, ss = rs - rr * 3;
+    const int rr = rs / fw, ss = rs - rr * fw;
     const int cin0 = k0 - rs * Cin;
     __attribute__((address_space(3))) char* la = lds3 + slot * SLOT;
     __attribute__((address_space(3))) char* lb = la + ABYTES;
     #pragma unroll
     for (int u = 0; u < ACHUNK; ++u) {
       int ih = a_ohS[u] + rr, iw = a_owS[u] + ss;
+      bool ok = a_n[u] >= 0 && ih >= 0 && iw >= 0;
+      if (D > 1) {  // dilated input: only multiples of D are stored rows/cols
+        ok = ok && (ih % D == 0) && (iw % D == 0);
+        ih /= D;
+        iw /= D;
+      }
       const char* src;
-      if (a_n[u] >= 0 && ih >= 0 && ih < H && iw >= 0 && iw < Wd) {
+      if (ok && ih < H && iw < Wd) {
         long pix = a_pix[u] + (long)ih * Wd + iw;
         src = (const char*)X + (pix * Cin + cin0) * 2 + a_col[u];
       } else {
@@ -184,39 +201,61 @@ __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
   }
 }
 
-extern "C" {
-
-void tfosr_conv3x3(const void* X, const void* W9, const void* guard, void* Y,
-                   int out_bf16, int N, int H, int W, int Cin, int Cout,
-                   int OH, int OW, int S, int P, hipStream_t s) {
+template <int D>
+static void launch_conv(const void* X, const void* W9, const void* guard,
+                        void* Y, int out_bf16, int N, int H, int W, int Cin,
+                        int Cout, int OH, int OW, int S, int P, int taps,
+                        int fw, hipStream_t s) {
   const long M = (long)N * OH * OW;
   if (Cout >= 192) {
     int ntm = (int)((M + 255) / 256), ntn = (Cout + 255) / 256;
     dim3 grid(ntm * ntn);
     if (out_bf16)
-      hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 2, 4, 8>), grid, dim3(512), 0,
-                         s, (const bf16_t*)X, (const bf16_t*)W9,
+      hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 2, 4, 8, D>), grid, dim3(512),
+                         0, s, (const bf16_t*)X, (const bf16_t*)W9,
                          (const bf16_t*)guard, (bf16_t*)Y, N, H, W, Cin, Cout,
-                         OH, OW, S, P);
+                         OH, OW, S, P, taps, fw);
     else
-      hipLaunchKernelGGL((conv3x3_kernel<float, 2, 4, 8>), grid, dim3(512), 0,
-                         s, (const bf16_t*)X, (const bf16_t*)W9,
+      hipLaunchKernelGGL((conv3x3_kernel<float, 2, 4, 8, D>), grid, dim3(512),
+                         0, s, (const bf16_t*)X, (const bf16_t*)W9,
                          (const bf16_t*)guard, (float*)Y, N, H, W, Cin, Cout,
-                         OH, OW, S, P);
+                         OH, OW, S, P, taps, fw);
     return;
   }
   int ntm = (int)((M + 255) / 256), ntn = (Cout + 127) / 128;
   dim3 grid(ntm * ntn);
   if (out_bf16)
-    hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 4, 2, 4>), grid, dim3(512), 0,
-                       s, (const bf16_t*)X, (const bf16_t*)W9,
+    hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 4, 2, 4, D>), grid, dim3(512),
+                       0, s, (const bf16_t*)X, (const bf16_t*)W9,
                        (const bf16_t*)guard, (bf16_t*)Y, N, H, W, Cin, Cout,
-                       OH, OW, S, P);
+                       OH, OW, S, P, taps, fw);
   else
-    hipLaunchKernelGGL((conv3x3_kernel<float, 4, 2, 4>), grid, dim3(512), 0,
-                       s, (const bf16_t*)X, (const bf16_t*)W9,
+    hipLaunchKernelGGL((conv3x3_kernel<float, 4, 2, 4, D>), grid, dim3(512),
+                       0, s, (const bf16_t*)X, (const bf16_t*)W9,
                        (const bf16_t*)guard, (float*)Y, N, H, W, Cin, Cout,
-                       OH, OW, S, P);
+                       OH, OW, S, P, taps, fw);
+}
+
+extern "C" {
+
+void tfosr_conv3x3(const void* X, const void* W9, const void* guard, void* Y,
+                   int out_bf16, int N, int H, int W, int Cin, int Cout,
+                   int OH, int OW, int S, int P, hipStream_t s) {
+  launch_conv<1>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH, OW, S, P,
+                 9, 3, s);
+}
+
+// General entry: taps = fh*fw; input dilation Dil in {1, 2}
+void tfosr_conv_mfma(const void* X, const void* W9, const void* guard, void* Y,
+                     int out_bf16, int N, int H, int W, int Cin, int Cout,
+                     int OH, int OW, int S, int P, int taps, int fw, int Dil,
+                     hipStream_t s) {
+  if (Dil == 2)
+    launch_conv<2>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH, OW, S, P,
+                   taps, fw, s);
+  else
+    launch_conv<1>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH, OW, S, P,
+                   taps, fw, s);
 }
 
 }  // extern "C"

@@ -20,16 +20,17 @@ from ..ops.modules import (Conv1x1, Conv3x3, FusedBN, FusedBNAddReLU,
 
 
 def conv3x3(cin, cout, stride=1):
-    if stride == 1 and cin % 32 == 0 and cout >= 64:
-        # stride-1 3x3s route to the implicit-GEMM MFMA kernel on gfx950
-        return Conv3x3(cin, cout)
+    if stride in (1, 2) and cin % 32 == 0 and cout % 32 == 0 and cout >= 64:
+        # 3x3s (stride 1 and 2) route to the implicit-GEMM MFMA kernel
+        return Conv3x3(cin, cout, stride)
     return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
 
 
 def conv1x1(cin, cout, stride=1):
-    if stride == 1:
-        # pointwise convs route to the hand-written MFMA GEMM on gfx950
-        return Conv1x1(cin, cout)
+    if stride == 1 or (stride == 2 and cin % 32 == 0 and cout % 32 == 0):
+        # pointwise convs route to the hand-written MFMA kernels on gfx950
+        # (stride 1: GEMM; stride 2: implicit-GEMM downsample conv)
+        return Conv1x1(cin, cout, stride)
     return nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
 
 

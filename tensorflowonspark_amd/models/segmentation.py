@@ -15,7 +15,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops.modules import FusedBNReLU
+from ..ops.modules import ConvTranspose2dMFMA, FusedBNReLU
 from .resnet import Bottleneck, ResNet
 
 
@@ -86,8 +86,9 @@ class UpBlock(nn.Module):
 
     def __init__(self, cin, cout):
         super().__init__()
-        self.up = nn.ConvTranspose2d(cin, cout, 4, stride=2, padding=1,
-                                     bias=False)
+        # routes to the dilated-input implicit-GEMM MFMA kernel on GPU when
+        # cin/cout are 32-multiples; falls back to the library conv otherwise
+        self.up = ConvTranspose2dMFMA(cin, cout, 4, stride=2, padding=1)
         self.bnrelu = FusedBNReLU(cout)
 
     def forward(self, x, skip=None):

@@ -18,7 +18,7 @@ import torch
 logger = logging.getLogger(__name__)
 
 #: must match TFOSR_API_VERSION in csrc/bindings.cpp
-API_VERSION = 3
+API_VERSION = 4
 
 _ext = None
 _ext_checked = False

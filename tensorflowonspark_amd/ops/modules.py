@@ -292,6 +292,43 @@ class FusedMaxPool2d(nn.Module):
 # 1x1 convolution as an MFMA GEMM (channels_last)
 # ---------------------------------------------------------------------------
 
+class _Conv1x1S2Fn(torch.autograd.Function):
+    """Stride-2 1x1 conv (ResNet downsample path) on the implicit-GEMM MFMA
+    kernel: forward is taps=1/S=2; backward-data is the input-dilated (D=2)
+    variant of the same kernel (dx = scatter of dy @ W, computed gather-side);
+    weight-grad reduces over the subsampled pixels via the library igemm until
+    conv_wrw covers stride 2."""
+
+    @staticmethod
+    def forward(ctx, x, weight):
+        ext = get_ext(required=True)
+        Cout, Cin = weight.shape[0], weight.shape[1]
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        w2d = weight.view(Cout, Cin).to(torch.bfloat16).contiguous()
+        y = ext.conv_mfma(x, w2d, Cout, 1, 1, 2, 0, 1, -1, -1)
+        ctx.save_for_backward(x, weight)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_ext(required=True)
+        x, weight = ctx.saved_tensors
+        Cout, Cin = weight.shape[0], weight.shape[1]
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        if dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
+        wT = weight.view(Cout, Cin).t().to(torch.bfloat16).contiguous()
+        # dx[h,w] = (h,w even) ? dy[h/2,w/2] @ W : 0  ==  D=2 dilated conv
+        dx = ext.conv_mfma(dy, wT, Cin, 1, 1, 1, 0, 2, x.shape[2], x.shape[3])
+        w4 = weight.to(torch.bfloat16).contiguous(
+            memory_format=torch.channels_last)
+        _, dw, _ = torch.ops.aten.convolution_backward(
+            dy, x, w4, None, [2, 2], [0, 0], [1, 1], False, [0, 0], 1,
+            [False, True, False])
+        return dx, dw.to(weight.dtype)
+
+
 class _Conv1x1Fn(torch.autograd.Function):
     """Stride-1 1x1 conv on channels_last tensors == GEMM over [N*H*W, Cin].
 
@@ -345,17 +382,21 @@ class _Conv1x1Fn(torch.autograd.Function):
 
 
 class Conv1x1(nn.Module):
-    """Pointwise convolution routed to the MFMA GEMM on GPU.
+    """Pointwise convolution routed to the MFMA GEMM on GPU (stride 1) or the
+    implicit-GEMM conv kernel (stride 2 — the ResNet downsample conv).
 
     Backend select via TFOS_CONV1X1: 'mfma' (default), 'blas'
     (torch.matmul / hipBLASLt for A/B comparison), 'miopen' (F.conv2d).
     Weight kept in Conv2d's [Cout, Cin, 1, 1] shape for state_dict parity.
     """
 
-    def __init__(self, cin, cout):
+    def __init__(self, cin, cout, stride=1):
         super().__init__()
+        self.stride = stride
         self.weight = nn.Parameter(torch.empty(cout, cin, 1, 1))
         nn.init.kaiming_normal_(self.weight, mode="fan_out", nonlinearity="relu")
+        # stride-2 backward-data needs K = Cout % 32 == 0 as well
+        self._s2_ok = stride == 2 and cin % 32 == 0 and cout % 32 == 0
 
     def forward(self, x):
         import os
@@ -364,6 +405,10 @@ class Conv1x1(nn.Module):
         # claimed-fp32 run, so other dtypes use the library conv
         if x.is_cuda and x.dtype == torch.bfloat16 and backend != "miopen":
             x = x.contiguous(memory_format=torch.channels_last)
+            if self.stride == 2:
+                if self._s2_ok and get_ext(required=True) is not None:
+                    return _Conv1x1S2Fn.apply(x, self.weight)
+                return F.conv2d(x, self.weight.to(x.dtype), stride=self.stride)
             if backend == "blas":
                 N, Cin, H, W = x.shape
                 x2d = x.permute(0, 2, 3, 1).reshape(-1, Cin)
@@ -374,11 +419,11 @@ class Conv1x1(nn.Module):
                 return y2d.view(N, H, W, -1).permute(0, 3, 1, 2)
             if get_ext(required=True) is not None:
                 return _Conv1x1Fn.apply(x, self.weight)
-        return F.conv2d(x, self.weight.to(x.dtype))
+        return F.conv2d(x, self.weight.to(x.dtype), stride=self.stride)
 
     def extra_repr(self):
-        return "{}x{} pointwise (MFMA)".format(self.weight.shape[1],
-                                               self.weight.shape[0])
+        return "{}x{} pointwise s{} (MFMA)".format(
+            self.weight.shape[1], self.weight.shape[0], self.stride)
 
 
 # ---------------------------------------------------------------------------
@@ -386,34 +431,39 @@ class Conv1x1(nn.Module):
 # ---------------------------------------------------------------------------
 
 class _Conv3x3Fn(torch.autograd.Function):
-    """Stride-1 3x3 conv on channels_last bf16 via the 4-deep-pipelined
-    implicit-GEMM kernel. dgrad = conv3x3 of dy with the flipped/transposed
-    weight (same kernel); wrw goes through MIOpen's tuned igemm."""
+    """3x3/pad-1 conv (stride 1 or 2) on channels_last bf16 via the
+    4-deep-pipelined implicit-GEMM kernel. dgrad = dilated-input conv of dy
+    with the flipped/transposed weight (same kernel, D=S); wrw goes through
+    MIOpen's tuned igemm by default (TFOS_WRW=mfma for the in-tree kernel)."""
 
     @staticmethod
-    def forward(ctx, x, weight):
+    def forward(ctx, x, weight, stride):
         ext = get_ext(required=True)
         Cout, Cin = weight.shape[0], weight.shape[1]
         if x.dtype != torch.bfloat16:
             x = x.to(torch.bfloat16)
         w9 = weight.permute(0, 2, 3, 1).reshape(Cout, 9 * Cin)             .to(torch.bfloat16).contiguous()
-        y = ext.conv3x3_fwd(x, w9, Cout, 1, 1)
+        y = ext.conv_mfma(x, w9, Cout, 3, 3, stride, 1, 1, -1, -1)
         ctx.save_for_backward(x, weight)
+        ctx.stride = stride
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = get_ext(required=True)
         x, weight = ctx.saved_tensors
+        S = ctx.stride
         Cout, Cin = weight.shape[0], weight.shape[1]
         dy = dy.contiguous(memory_format=torch.channels_last)
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
         # W'[cin][r][s][cout] = W[cout][2-r][2-s][cin]
         w9p = weight.flip(2, 3).permute(1, 2, 3, 0)             .reshape(Cin, 9 * Cout).to(torch.bfloat16).contiguous()
-        dx = ext.conv3x3_fwd(dy, w9p, Cin, 1, 1)
+        # dx = conv(dy input-dilated by S, W'), pad' = (3-1) - pad = 1
+        dx = ext.conv_mfma(dy, w9p, Cin, 3, 3, 1, 1, S,
+                           x.shape[2], x.shape[3])
         import os
-        if os.environ.get("TFOS_WRW", "miopen") == "mfma":
+        if os.environ.get("TFOS_WRW", "miopen") == "mfma" and S == 1:
             # dW9[cout][r][s][cin] -> [Cout, Cin, 3, 3]
             dw9 = ext.conv_wrw(dy, x, 3, 3, 1)
             dw = dw9.view(Cout, 3, 3, Cin).permute(0, 3, 1, 2).contiguous()
@@ -421,35 +471,123 @@ class _Conv3x3Fn(torch.autograd.Function):
             w4 = weight.to(torch.bfloat16).contiguous(
                 memory_format=torch.channels_last)
             _, dw, _ = torch.ops.aten.convolution_backward(
-                dy, x, w4, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+                dy, x, w4, None, [S, S], [1, 1], [1, 1], False, [0, 0], 1,
                 [False, True, False])
-        return dx, dw.to(weight.dtype)
+        return dx, dw.to(weight.dtype), None
 
 
 class Conv3x3(nn.Module):
-    """3x3/stride-1/pad-1 conv routed to the implicit-GEMM MFMA kernel.
+    """3x3/pad-1 conv (stride 1 or 2) routed to the implicit-GEMM MFMA kernel.
 
     Backend select via TFOS_CONV3X3: 'mfma' (default), 'miopen'.
     Weight in Conv2d's [Cout, Cin, 3, 3] shape for state_dict parity.
     """
 
-    def __init__(self, cin, cout):
+    def __init__(self, cin, cout, stride=1):
         super().__init__()
+        self.stride = stride
         self.weight = nn.Parameter(torch.empty(cout, cin, 3, 3))
         nn.init.kaiming_normal_(self.weight, mode="fan_out", nonlinearity="relu")
-        self._eligible = cin % 32 == 0 and cout % 32 == 0 and cout >= 64
+        self._eligible = cin % 32 == 0 and cout % 32 == 0 and cout >= 64 \
+            and stride in (1, 2)
 
     def forward(self, x):
         import os
         backend = os.environ.get("TFOS_CONV3X3", "mfma")
-        if x.is_cuda and self._eligible and backend == "mfma"                 and get_ext(required=True) is not None:
+        if x.is_cuda and self._eligible and backend == "mfma"                 and x.dtype == torch.bfloat16                 and get_ext(required=True) is not None:
             x = x.contiguous(memory_format=torch.channels_last)
-            return _Conv3x3Fn.apply(x, self.weight)
-        return F.conv2d(x, self.weight.to(x.dtype), padding=1)
+            return _Conv3x3Fn.apply(x, self.weight, self.stride)
+        return F.conv2d(x, self.weight.to(x.dtype), padding=1,
+                        stride=self.stride)
 
     def extra_repr(self):
-        return "{}x{} 3x3 (implicit-GEMM MFMA)".format(self.weight.shape[1],
-                                                       self.weight.shape[0])
+        return "{}x{} 3x3 s{} (implicit-GEMM MFMA)".format(
+            self.weight.shape[1], self.weight.shape[0], self.stride)
+
+
+# ---------------------------------------------------------------------------
+# Transposed convolution (U-Net/DeepLab decoders) as a dilated-input conv
+# ---------------------------------------------------------------------------
+
+class _ConvT2dFn(torch.autograd.Function):
+    """ConvTranspose2d (stride 2) == conv over the zero-dilated input:
+    y = conv(x_dil(D=2), W^swap-flip, S=1, P'=k-1-P). Backward-data is the
+    plain stride-2 conv of dy with the unswapped weight — both directions run
+    on the same implicit-GEMM MFMA kernel (reference workload: pix2pix
+    upsample k4 s2 + Conv2DTranspose k3 s2, segmentation_spark.py:85-97)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, stride, padding, output_padding):
+        ext = get_ext(required=True)
+        Cin, Cout, KH, KW = weight.shape
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        # W'[cout][r][s][cin] = W[cin][cout][KH-1-r][KW-1-s]
+        wk = weight.flip(2, 3).permute(1, 2, 3, 0)             .reshape(Cout, KH * KW * Cin).to(torch.bfloat16).contiguous()
+        H, W = x.shape[2], x.shape[3]
+        OH = (H - 1) * stride - 2 * padding + KH + output_padding
+        OW = (W - 1) * stride - 2 * padding + KW + output_padding
+        y = ext.conv_mfma(x, wk, Cout, KH, KW, 1, KH - 1 - padding, stride,
+                          OH, OW)
+        ctx.save_for_backward(x, weight)
+        ctx.params = (stride, padding)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_ext(required=True)
+        x, weight = ctx.saved_tensors
+        S, P = ctx.params
+        Cin, Cout, KH, KW = weight.shape
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        if dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
+        # dx = plain stride-S conv of dy with W (no flip/swap)
+        wd = weight.permute(0, 2, 3, 1).reshape(Cin, KH * KW * Cout)             .to(torch.bfloat16).contiguous()
+        dx = ext.conv_mfma(dy, wd, Cin, KH, KW, S, P, 1,
+                           x.shape[2], x.shape[3])
+        w4 = weight.to(torch.bfloat16).contiguous(
+            memory_format=torch.channels_last)
+        _, dw, _ = torch.ops.aten.convolution_backward(
+            dy, x, w4, None, [S, S], [P, P], [1, 1], True, [0, 0], 1,
+            [False, True, False])
+        return dx, dw.to(weight.dtype), None, None, None
+
+
+class ConvTranspose2dMFMA(nn.Module):
+    """Transposed conv routed to the implicit-GEMM MFMA kernel on GPU.
+
+    Weight kept in nn.ConvTranspose2d's [Cin, Cout, K, K] layout for
+    state_dict parity; bias unsupported (decoders here use BN right after).
+    """
+
+    def __init__(self, cin, cout, kernel_size=4, stride=2, padding=1,
+                 output_padding=0):
+        super().__init__()
+        self.stride, self.padding = stride, padding
+        self.output_padding = output_padding
+        self.weight = nn.Parameter(torch.empty(cin, cout, kernel_size,
+                                               kernel_size))
+        nn.init.kaiming_normal_(self.weight, mode="fan_in", nonlinearity="relu")
+        # fwd K = taps*Cin, dgrad K = taps*Cout — both must be 32-multiples
+        self._eligible = cin % 32 == 0 and cout % 32 == 0 and stride == 2
+
+    def forward(self, x):
+        import os
+        if x.is_cuda and self._eligible and x.dtype == torch.bfloat16 \
+                and os.environ.get("TFOS_CONVT", "mfma") == "mfma" \
+                and get_ext(required=True) is not None:
+            x = x.contiguous(memory_format=torch.channels_last)
+            return _ConvT2dFn.apply(x, self.weight, self.stride, self.padding,
+                                    self.output_padding)
+        return F.conv_transpose2d(x, self.weight.to(x.dtype),
+                                  stride=self.stride, padding=self.padding,
+                                  output_padding=self.output_padding)
+
+    def extra_repr(self):
+        return "{}->{} k{} s{} p{} (implicit-GEMM MFMA)".format(
+            self.weight.shape[0], self.weight.shape[1], self.weight.shape[2],
+            self.stride, self.padding)
 
 
 # ---------------------------------------------------------------------------
@@ -467,6 +605,88 @@ def gemm_bf16(a, b):
         if ext is not None:
             return ext.gemm_bf16(a.contiguous(), b.contiguous())
     return (a.float() @ b.float())
+
+
+class _GemmBTFn(torch.autograd.Function):
+    """Autograd-composable C[M,N] = A[M,K] @ B[N,K]^T on the MFMA kernel.
+
+    dA re-runs the same kernel (dA = dC @ B, K-contiguous after a small
+    transpose of B); dB — a contraction over the M dim — uses the library
+    GEMM per the kernel-usage policy."""
+
+    @staticmethod
+    def forward(ctx, a, b):
+        ext = get_ext(required=True)
+        a = a.to(torch.bfloat16).contiguous()
+        b = b.to(torch.bfloat16).contiguous()
+        ctx.save_for_backward(a, b)
+        return ext.gemm_bt(a, b, True)
+
+    @staticmethod
+    def backward(ctx, dc):
+        ext = get_ext(required=True)
+        a, b = ctx.saved_tensors
+        dc = dc.to(torch.bfloat16).contiguous()
+        da = ext.gemm_bt(dc, b.t().contiguous(), True)
+        db = (dc.float().t() @ a.float()).to(b.dtype)
+        return da, db
+
+
+class DenseMFMA(nn.Module):
+    """Dense/Linear layer on the MFMA GEMM (survey §2.3 GEMM/Dense row:
+    reference ``Dense(64, relu)``/``Dense(10)``, ``mnist_spark.py:17-19``).
+    Weight/bias layout matches nn.Linear for state_dict parity."""
+
+    def __init__(self, in_features, out_features, bias=True):
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(out_features, in_features))
+        self.bias = nn.Parameter(torch.zeros(out_features)) if bias else None
+        nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
+
+    def forward(self, x):
+        if x.is_cuda and x.dtype == torch.bfloat16 \
+                and self.weight.shape[1] % 32 == 0 \
+                and get_ext(required=True) is not None:
+            y = _GemmBTFn.apply(x, self.weight)
+            if self.bias is not None:
+                y = y + self.bias.to(y.dtype)
+            return y
+        return F.linear(x, self.weight.to(x.dtype),
+                        None if self.bias is None else self.bias.to(x.dtype))
+
+
+class Conv2dIm2colMFMA(nn.Module):
+    """Small-Cin conv (e.g. MNIST's Conv2D(1->32, k3)) via im2col into the
+    MFMA GEMM, K zero-padded to a 32-multiple. The unfold/pad are tiny at
+    MNIST scale; the matmul — the hot part — runs on the in-tree kernel.
+    Weight/bias layout matches nn.Conv2d."""
+
+    def __init__(self, cin, cout, kernel_size, bias=True):
+        super().__init__()
+        self.k = kernel_size
+        self.weight = nn.Parameter(torch.empty(cout, cin, kernel_size,
+                                               kernel_size))
+        self.bias = nn.Parameter(torch.zeros(cout)) if bias else None
+        nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
+
+    def forward(self, x):
+        if x.is_cuda and x.dtype == torch.bfloat16 \
+                and get_ext(required=True) is not None:
+            N, Cin, H, W = x.shape
+            Cout = self.weight.shape[0]
+            OH, OW = H - self.k + 1, W - self.k + 1
+            K = Cin * self.k * self.k
+            Kpad = (K + 31) // 32 * 32
+            cols = F.unfold(x, self.k)                    # [N, K, L]
+            a = cols.transpose(1, 2).reshape(-1, K)       # [N*L, K]
+            a = F.pad(a, (0, Kpad - K))
+            w2 = F.pad(self.weight.view(Cout, K), (0, Kpad - K))
+            y2 = _GemmBTFn.apply(a, w2)                   # [N*L, Cout]
+            if self.bias is not None:
+                y2 = y2 + self.bias.to(y2.dtype)
+            return y2.view(N, OH * OW, Cout).transpose(1, 2)                 .reshape(N, Cout, OH, OW)
+        b = None if self.bias is None else self.bias.to(x.dtype)
+        return F.conv2d(x, self.weight.to(x.dtype), b)
 
 
 class BucketAdam:
