@@ -47,6 +47,10 @@ void tfosr_conv_mfma(const void*, const void*, const void*, void*, int, int,
                      hipStream_t);
 void tfosr_conv_wrw(const void*, const void*, float*, int, int, int, int, int,
                     int, int, int, int, int, hipStream_t);
+int tfosr_wrw2_split(int, int, long);
+void tfosr_conv_wrw2(const void*, const void*, const void*, float*, float*,
+                     int, int, int, int, int, int, int, int, int, int, int,
+                     int, hipStream_t);
 }
 
 namespace tfosr {
@@ -375,6 +379,33 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                     y.data_ptr(), /*out_bf16=*/1, N, H, W, Cin, Cout, OH, OW,
                     S, P, taps, fw, D, cur_stream());
     return y;
+  });
+  // wrw v2: MFMA + hardware transpose-reads + split-M workspace.
+  // dy [N,Cout,OH,OW] cl, x [N,Cin,H,W] cl; filter R x S_f, conv stride,
+  // pad P. Returns fp32 dW [Cout, R*S_f*Cin].
+  m.def("conv_wrw2", [](at::Tensor dy, at::Tensor x, long R, long S_f,
+                        long stride, long P) {
+    TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                x.is_contiguous(at::MemoryFormat::ChannelsLast));
+    TORCH_CHECK(dy.scalar_type() == at::kBFloat16 &&
+                x.scalar_type() == at::kBFloat16);
+    int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+    int Cout = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+    TORCH_CHECK(Cin % 8 == 0 && Cout % 8 == 0,
+                "conv_wrw2 needs channel counts % 8 == 0");
+    long taps = R * S_f;
+    TORCH_CHECK(taps == 1 || taps == 9, "conv_wrw2 supports 1x1 and 3x3");
+    long M = (long)N * OH * OW;
+    int split = tfosr_wrw2_split(Cout, Cin, M);
+    long K = taps * (long)Cin;
+    auto ws = at::empty({(long)split, (long)Cout * K},
+                        x.options().dtype(at::kFloat));
+    auto dW = at::empty({(long)Cout, K}, x.options().dtype(at::kFloat));
+    auto guard = at::zeros({64}, x.options());
+    tfosr_conv_wrw2(dy.data_ptr(), x.data_ptr(), guard.data_ptr(),
+                    ws.data_ptr<float>(), dW.data_ptr<float>(), N, H, W, Cin,
+                    Cout, OH, OW, R, S_f, stride, P, split, cur_stream());
+    return dW;
   });
   m.def("conv_wrw", [](at::Tensor dy, at::Tensor x, long R, long S, long P) {
     TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast) &&

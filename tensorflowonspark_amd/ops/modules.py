@@ -321,11 +321,16 @@ class _Conv1x1S2Fn(torch.autograd.Function):
         wT = weight.view(Cout, Cin).t().to(torch.bfloat16).contiguous()
         # dx[h,w] = (h,w even) ? dy[h/2,w/2] @ W : 0  ==  D=2 dilated conv
         dx = ext.conv_mfma(dy, wT, Cin, 1, 1, 1, 0, 2, x.shape[2], x.shape[3])
-        w4 = weight.to(torch.bfloat16).contiguous(
-            memory_format=torch.channels_last)
-        _, dw, _ = torch.ops.aten.convolution_backward(
-            dy, x, w4, None, [2, 2], [0, 0], [1, 1], False, [0, 0], 1,
-            [False, True, False])
+        import os
+        if os.environ.get("TFOS_WRW", "mfma2") == "mfma2" \
+                and Cin % 8 == 0 and Cout % 8 == 0:
+            dw = ext.conv_wrw2(dy, x, 1, 1, 2, 0).view(Cout, Cin, 1, 1)
+        else:
+            w4 = weight.to(torch.bfloat16).contiguous(
+                memory_format=torch.channels_last)
+            _, dw, _ = torch.ops.aten.convolution_backward(
+                dy, x, w4, None, [2, 2], [0, 0], [1, 1], False, [0, 0], 1,
+                [False, True, False])
         return dx, dw.to(weight.dtype)
 
 
@@ -365,12 +370,15 @@ class _Conv1x1Fn(torch.autograd.Function):
         dx2d = ext.gemm_bt(dy2d, w2d.t().contiguous(), True)
         dx = dx2d.view(N, H, W, Cin).permute(0, 3, 1, 2)
         # dW[Cout,Cin] = dy^T @ x: a reduction over the huge M dim.
-        # TFOS_WRW=mfma -> our TN kernel; default MIOpen wrw igemm
-        # (hipBLASLt's split-K pick was ~7x slower than either).
+        # TFOS_WRW: mfma2 (default, transpose-read MFMA kernel), mfma
+        # (round-1 TN kernel), miopen (library igemm; hipBLASLt's split-K
+        # pick was ~7x slower than either).
         import os
+        wrw = os.environ.get("TFOS_WRW", "mfma2")
         x4d = x2d.view(N, H, W, Cin).permute(0, 3, 1, 2)
-        if os.environ.get("TFOS_WRW", "miopen") == "mfma" and Cin % 8 == 0 \
-                and Cout % 8 == 0:
+        if wrw == "mfma2" and Cin % 8 == 0 and Cout % 8 == 0:
+            dw = ext.conv_wrw2(dy, x4d, 1, 1, 1, 0).view(Cout, Cin, 1, 1)
+        elif wrw == "mfma" and Cin % 8 == 0 and Cout % 8 == 0:
             ext2 = get_ext(required=True)
             dw = ext2.conv_wrw(dy, x4d, 1, 1, 0).view(Cout, Cin, 1, 1)
         else:
@@ -463,8 +471,13 @@ class _Conv3x3Fn(torch.autograd.Function):
         dx = ext.conv_mfma(dy, w9p, Cin, 3, 3, 1, 1, S,
                            x.shape[2], x.shape[3])
         import os
-        if os.environ.get("TFOS_WRW", "miopen") == "mfma" and S == 1:
-            # dW9[cout][r][s][cin] -> [Cout, Cin, 3, 3]
+        wrw = os.environ.get("TFOS_WRW", "mfma2")
+        if wrw == "mfma2" and Cin % 8 == 0 and Cout % 8 == 0:
+            # transpose-read MFMA wrw kernel (stride 1 and 2)
+            dw9 = ext.conv_wrw2(dy, x, 3, 3, S, 1)
+            dw = dw9.view(Cout, 3, 3, Cin).permute(0, 3, 1, 2).contiguous()
+        elif wrw == "mfma" and S == 1:
+            # round-1 TN kernel (kept for comparison)
             dw9 = ext.conv_wrw(dy, x, 3, 3, 1)
             dw = dw9.view(Cout, 3, 3, Cin).permute(0, 3, 1, 2).contiguous()
         else:

@@ -165,3 +165,32 @@ def test_mnist_cnn_on_hip_path():
     _close(y, yr, 5e-2, 3e-1, "mnist fwd")
     _close(m.features[0].weight.grad, mc.features[0].weight.grad,
            5e-2, 3e-1, "mnist conv dw")
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("case", [
+    (4, 64, 16, 16, 128, 3, 1, 1),    # 3x3 s1
+    (2, 128, 28, 28, 128, 3, 2, 1),   # 3x3 s2
+    (3, 72, 13, 9, 88, 3, 1, 1),      # odd spatial, ragged channels (x8)
+    (4, 256, 14, 14, 512, 1, 1, 0),   # 1x1 s1
+    (2, 256, 28, 28, 512, 1, 2, 0),   # 1x1 s2 (downsample)
+])
+def test_conv_wrw2_kernel(case):
+    """wrw v2 (transpose-read MFMA + split-M workspace) vs fp32 torch."""
+    from tensorflowonspark_amd.ops import get_ext
+    ext = get_ext(required=True)
+    torch.manual_seed(7)
+    N, Cin, H, W, Cout, k, s, p = case
+    x = torch.randn(N, Cin, H, W, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    OH = (H + 2 * p - k) // s + 1
+    OW = (W + 2 * p - k) // s + 1
+    dy = torch.randn(N, Cout, OH, OW, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    dw = ext.conv_wrw2(dy, x, k, k, s, p)          # [Cout, k*k*Cin]
+    w_ref = torch.zeros(Cout, Cin, k, k, requires_grad=True)
+    y = F.conv2d(x.float().cpu(), w_ref, stride=s, padding=p)
+    y.backward(dy.float().cpu())
+    ref = w_ref.grad.permute(0, 2, 3, 1).reshape(Cout, k * k * Cin)
+    _close(dw, ref, 2e-2, 2e-1 * max(1, N // 2), "wrw2")

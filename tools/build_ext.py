@@ -20,7 +20,7 @@ OUT_SO = os.path.join(OUT_DIR, "tfosr_hip_ops.so")
 BUILD = os.path.join(REPO, "build", "hip")
 
 SOURCES = ["bn_relu.hip", "softmax_xent.hip", "elementwise.hip",
-           "gemm_mfma.hip", "conv3x3_mfma.hip", "conv_wrw_mfma.hip", "maxpool.hip", "tfrecord_codec.cpp", "bindings.cpp"]
+           "gemm_mfma.hip", "conv3x3_mfma.hip", "conv_wrw_mfma.hip", "conv_wrw2.hip", "maxpool.hip", "tfrecord_codec.cpp", "bindings.cpp"]
 
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
