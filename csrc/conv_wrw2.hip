@@ -8,13 +8,14 @@
 // (commit 845fb1f). This kernel instead:
 //
 //   * stages dy / per-tap x tiles with coalesced 16 B global_load_lds chunks
-//     into a PERMUTED layout f(m, c) = c&15 + (m&3)*16 + (m>>3)*64 +
-//     ((m>>2)&1)*256 (per 16-channel block) — chosen so that gfx950's
-//     hardware transpose-read ds_read_b64_tr_b16 (lane l elem j reads
-//     lds[(l&15) + j*16 + (l>>4)*64], guide T10) delivers exactly the
-//     k(=m)-major MFMA fragment: lane l gets m = (l>>4)*8 + t*4 + j for
-//     channel l&15, t = the offset:512 immediate. Chunks stay 16 B
-//     contiguous because channels are the fast axis in both layouts.
+//     into plain row-major [32 m][16 c] blocks, then uses gfx950's hardware
+//     transpose-read ds_read_b64_tr_b16 to pull k(=m)-major MFMA fragments.
+//     Measured semantics (tools/trprobe.hip on MI355X): dest[l][j] =
+//     lds[align8B(addr_of_lane((l&48) + 4j + ((l>>2)&3))) + (l&3) elems].
+//     With per-lane address ((l>>4)*8 + 4t + ((l&15)>>2))*16 + (l&3)*4
+//     elements into the row-major block, lane l receives exactly
+//     A[channel = l&15][m = (l>>4)*8 + 4t + j] — the MFMA operand — while
+//     staging stays 16 B contiguous (channels are the fast axis in HBM).
 //   * split-M: each workgroup owns a contiguous m-range and writes its
 //     partial tile to a workspace slice (plain stores); a tiny reduce kernel
 //     sums the slices — no global atomics.
@@ -80,17 +81,17 @@ __global__ __launch_bounds__(512, 1) void conv_wrw2_kernel(
   // per-thread chunk descriptors (fixed across k-steps except the m part)
   int ch_region[CPT], ch_mloc[CPT], ch_ldsoff[CPT];
   long ch_coff[CPT];   // channel byte offset into the global row; <0 => guard
+  // global_load_lds writes lane-linearly from the wave base, so the LDS
+  // destination is the plain linear chunk index; the (region, block, m, h)
+  // the chunk HOLDS is decoded from that index instead.
   #pragma unroll
   for (int u = 0; u < CPT; ++u) {
-    int d = t + u * 512;
-    int region = d >> 8;
-    int cid = d & 255;
-    int m = cid >> 3, cb = (cid >> 1) & 3, h = cid & 1;
+    int q = t + u * 512;               // linear 16 B chunk index
+    int region = q >> 8;
+    int m = (q >> 1) & 31, cb = (q >> 6) & 3, h = q & 1;
     ch_region[u] = region;
     ch_mloc[u] = m;
-    ch_ldsoff[u] = region * REGB +
-        (cb * 512 + h * 8 + (m & 3) * 16 + ((m >> 2) & 1) * 256 +
-         (m >> 3) * 64) * 2;
+    ch_ldsoff[u] = q * 16;
     int cglob = (region == 0 ? cout0 : cin0) + cb * 16 + h * 8;
     int climit = region == 0 ? Cout : Cin;
     ch_coff[u] = (cglob + 7 < climit) ? (long)cglob * 2 : -1;
@@ -134,15 +135,17 @@ __global__ __launch_bounds__(512, 1) void conv_wrw2_kernel(
   if (nkt > steps_per_wg) nkt = steps_per_wg;
   if (nkt <= 0) nkt = 0;
 
-  // lane part of the transpose-read address (elements)
-  const int lbase = (lane & 15) + (lane >> 4) * 64;
+  // per-lane transpose-read address into a row-major [32m][16c] block
+  // (bytes); the t=1 half (m += 4) sits +128 B further
+  const int lbase =
+      ((((lane >> 4) * 8 + ((lane & 15) >> 2)) * 16) + (lane & 3) * 4) * 2;
 
   auto trread8 = [&](AS3 char* region, int blk) -> bf16x8 {
     AS3 s16x4* p =
-        (AS3 s16x4*)(region + blk * 1024 + lbase * 2);
+        (AS3 s16x4*)(region + blk * 1024 + lbase);
     s16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(p);
     s16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-        (AS3 s16x4*)((AS3 char*)p + 512));
+        (AS3 s16x4*)((AS3 char*)p + 128));
     bf16x8 f;
     f[0] = lo[0]; f[1] = lo[1]; f[2] = lo[2]; f[3] = lo[3];
     f[4] = hi[0]; f[5] = hi[1]; f[6] = hi[2]; f[7] = hi[3];
