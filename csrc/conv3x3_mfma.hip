@@ -154,7 +154,10 @@ __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
   constexpr int INFLIGHT = 2 * (ACHUNK + BCHUNK);  // 2 staged tiles in flight
 
   for (int kt = 0; kt < nkt; ++kt) {
-    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(INFLIGHT) : "memory");
+    if (kt + 2 < nkt)
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(INFLIGHT) : "memory");
+    else  // tail: fewer newer slots in flight than the counted wait assumes
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_sched_barrier(0);
     __syncthreads();
     if (kt + 3 < nkt) stage_tile((kt + 3) & 3, kt + 3);

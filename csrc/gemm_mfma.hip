@@ -250,7 +250,10 @@ __global__ __launch_bounds__(512, 1) void gemm_bt256_kernel(
   const int kslot = lane >> 4;
 
   for (int kt = 0; kt < nkt; ++kt) {
-    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    if (kt + 2 < nkt)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else  // tail: fewer newer tiles in flight than the counted wait assumes
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_sched_barrier(0);
     __syncthreads();
     if (kt + 3 < nkt) stage_tile((kt + 3) & 3, kt + 3);

@@ -34,6 +34,9 @@ int main(int argc, char** argv) {
   if (argc > 2) Cout = atoi(argv[2]);
   if (argc > 3) Cin = atoi(argv[3]);
   if (argc > 4) stride = atoi(argv[4]);
+  if (argc > 5) { H = atoi(argv[5]); W = H; }
+  if (argc > 6) N = atoi(argv[6]);
+  if (argc > 7) W = atoi(argv[7]);
   if (K == 3) P = 1;
   int OH = (H + 2 * P - K) / stride + 1, OW = (W + 2 * P - K) / stride + 1;
   long M = (long)N * OH * OW;
