@@ -35,6 +35,8 @@ void tfosr_adam_step(float*, const float*, float*, float*, float, float, float,
                      float, float, int, int, long, hipStream_t);
 void tfosr_gemm_bt(const void*, const void*, void*, int, int, int, int,
                    hipStream_t);
+void tfosr_gemm_bt_acc(const void*, const void*, void*, int, int, int, int,
+                       int, hipStream_t);
 void tfosr_mfma_probe(const short*, const short*, float*, hipStream_t);
 void tfosr_maxpool_fwd(const void*, void*, unsigned char*, int, int, int, int,
                        int, int, int, int, int, int, hipStream_t);
@@ -44,7 +46,7 @@ void tfosr_conv3x3(const void*, const void*, const void*, void*, int, int, int,
                    int, int, int, int, int, int, int, hipStream_t);
 void tfosr_conv_mfma(const void*, const void*, const void*, void*, int, int,
                      int, int, int, int, int, int, int, int, int, int, int,
-                     hipStream_t);
+                     int, hipStream_t);
 void tfosr_conv_wrw(const void*, const void*, float*, int, int, int, int, int,
                     int, int, int, int, int, hipStream_t);
 int tfosr_wrw2_split(int, int, long);
@@ -377,8 +379,44 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     auto guard = at::zeros({64}, x.options());
     tfosr_conv_mfma(x.data_ptr(), wk.contiguous().data_ptr(), guard.data_ptr(),
                     y.data_ptr(), /*out_bf16=*/1, N, H, W, Cin, Cout, OH, OW,
-                    S, P, taps, fw, D, cur_stream());
+                    S, P, taps, fw, D, 0, cur_stream());
     return y;
+  });
+  // conv_mfma accumulating INTO an existing channels_last tensor:
+  // out += conv(x, wk) — fuses the residual-join gradient accumulation the
+  // eager autograd would do as a separate whole-tensor add (8.4 ms/step of
+  // CUDAFunctor_add at ResNet-50 b1024, profiles/README r01).
+  m.def("conv_mfma_acc", [](at::Tensor x, at::Tensor wk, at::Tensor out,
+                            long fh, long fw, long S, long P, long D) {
+    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                out.is_contiguous(at::MemoryFormat::ChannelsLast));
+    TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+                wk.scalar_type() == at::kBFloat16 &&
+                out.scalar_type() == at::kBFloat16);
+    int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+    TORCH_CHECK(Cin % 32 == 0, "conv_mfma requires Cin % 32 == 0");
+    long taps = fh * fw;
+    TORCH_CHECK(wk.size(1) == taps * Cin);
+    long Cout = out.size(1), OH = out.size(2), OW = out.size(3);
+    TORCH_CHECK(out.size(0) == N);
+    auto guard = at::zeros({64}, x.options());
+    tfosr_conv_mfma(x.data_ptr(), wk.contiguous().data_ptr(), guard.data_ptr(),
+                    out.data_ptr(), /*out_bf16=*/1, N, H, W, Cin, Cout, OH, OW,
+                    S, P, taps, fw, D, 1, cur_stream());
+    return out;
+  });
+  // gemm accumulating into an existing [M, N] tensor: C += A @ B^T
+  m.def("gemm_bt_acc", [](at::Tensor a, at::Tensor b, at::Tensor c) {
+    TORCH_CHECK(a.is_contiguous() && b.is_contiguous() && c.is_contiguous());
+    TORCH_CHECK(a.scalar_type() == at::kBFloat16 &&
+                b.scalar_type() == at::kBFloat16);
+    long M = a.size(0), K = a.size(1), Nn = b.size(0);
+    TORCH_CHECK(b.size(1) == K && c.size(0) == M && c.size(1) == Nn);
+    TORCH_CHECK(K % 32 == 0);
+    tfosr_gemm_bt_acc(a.data_ptr(), b.data_ptr(), c.data_ptr(),
+                      c.scalar_type() == at::kBFloat16 ? 1 : 0, M, Nn, K, 1,
+                      cur_stream());
+    return c;
   });
   // wrw v2: MFMA + hardware transpose-reads + split-M workspace.
   // dy [N,Cout,OH,OW] cl, x [N,Cin,H,W] cl; filter R x S_f, conv stride,

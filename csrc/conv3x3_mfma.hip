@@ -39,7 +39,7 @@ __device__ __forceinline__ void c3_stage16(const char* src,
 }
 
 // WRG x WCG wave grid (8 waves); per-wave output (MI*16) x 64
-template <typename OT, int WRG, int WCG, int MI, int D = 1>
+template <typename OT, int WRG, int WCG, int MI, int D = 1, bool ACC = false>
 __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
     const bf16_t* __restrict__ X, const bf16_t* __restrict__ W9,
     const bf16_t* __restrict__ guard, OT* __restrict__ C,
@@ -198,13 +198,16 @@ __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
       #pragma unroll
       for (int nj = 0; nj < 4; ++nj) {
         long n = tile_n + wc * 64 + nj * 16 + ccol;
-        if (n < Cout) C[m * Cout + n] = (OT)acc[mi][nj][r];
+        if (n < Cout)
+          C[m * Cout + n] = ACC
+              ? (OT)((float)C[m * Cout + n] + acc[mi][nj][r])
+              : (OT)acc[mi][nj][r];
       }
     }
   }
 }
 
-template <int D>
+template <int D, bool ACC>
 static void launch_conv(const void* X, const void* W9, const void* guard,
                         void* Y, int out_bf16, int N, int H, int W, int Cin,
                         int Cout, int OH, int OW, int S, int P, int taps,
@@ -214,13 +217,13 @@ static void launch_conv(const void* X, const void* W9, const void* guard,
     int ntm = (int)((M + 255) / 256), ntn = (Cout + 255) / 256;
     dim3 grid(ntm * ntn);
     if (out_bf16)
-      hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 2, 4, 8, D>), grid, dim3(512),
-                         0, s, (const bf16_t*)X, (const bf16_t*)W9,
+      hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 2, 4, 8, D, ACC>), grid,
+                         dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)W9,
                          (const bf16_t*)guard, (bf16_t*)Y, N, H, W, Cin, Cout,
                          OH, OW, S, P, taps, fw);
     else
-      hipLaunchKernelGGL((conv3x3_kernel<float, 2, 4, 8, D>), grid, dim3(512),
-                         0, s, (const bf16_t*)X, (const bf16_t*)W9,
+      hipLaunchKernelGGL((conv3x3_kernel<float, 2, 4, 8, D, ACC>), grid,
+                         dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)W9,
                          (const bf16_t*)guard, (float*)Y, N, H, W, Cin, Cout,
                          OH, OW, S, P, taps, fw);
     return;
@@ -228,13 +231,13 @@ static void launch_conv(const void* X, const void* W9, const void* guard,
   int ntm = (int)((M + 255) / 256), ntn = (Cout + 127) / 128;
   dim3 grid(ntm * ntn);
   if (out_bf16)
-    hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 4, 2, 4, D>), grid, dim3(512),
-                       0, s, (const bf16_t*)X, (const bf16_t*)W9,
+    hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 4, 2, 4, D, ACC>), grid,
+                       dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)W9,
                        (const bf16_t*)guard, (bf16_t*)Y, N, H, W, Cin, Cout,
                        OH, OW, S, P, taps, fw);
   else
-    hipLaunchKernelGGL((conv3x3_kernel<float, 4, 2, 4, D>), grid, dim3(512),
-                       0, s, (const bf16_t*)X, (const bf16_t*)W9,
+    hipLaunchKernelGGL((conv3x3_kernel<float, 4, 2, 4, D, ACC>), grid,
+                       dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)W9,
                        (const bf16_t*)guard, (float*)Y, N, H, W, Cin, Cout,
                        OH, OW, S, P, taps, fw);
 }
@@ -244,21 +247,30 @@ extern "C" {
 void tfosr_conv3x3(const void* X, const void* W9, const void* guard, void* Y,
                    int out_bf16, int N, int H, int W, int Cin, int Cout,
                    int OH, int OW, int S, int P, hipStream_t s) {
-  launch_conv<1>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH, OW, S, P,
-                 9, 3, s);
+  launch_conv<1, false>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH, OW,
+                        S, P, 9, 3, s);
 }
 
 // General entry: taps = fh*fw; input dilation Dil in {1, 2}
 void tfosr_conv_mfma(const void* X, const void* W9, const void* guard, void* Y,
                      int out_bf16, int N, int H, int W, int Cin, int Cout,
                      int OH, int OW, int S, int P, int taps, int fw, int Dil,
-                     hipStream_t s) {
-  if (Dil == 2)
-    launch_conv<2>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH, OW, S, P,
-                   taps, fw, s);
-  else
-    launch_conv<1>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH, OW, S, P,
-                   taps, fw, s);
+                     int accum, hipStream_t s) {
+  if (Dil == 2) {
+    if (accum)
+      launch_conv<2, true>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH,
+                           OW, S, P, taps, fw, s);
+    else
+      launch_conv<2, false>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH,
+                            OW, S, P, taps, fw, s);
+  } else {
+    if (accum)
+      launch_conv<1, true>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH,
+                           OW, S, P, taps, fw, s);
+    else
+      launch_conv<1, false>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH,
+                            OW, S, P, taps, fw, s);
+  }
 }
 
 }  // extern "C"

@@ -39,7 +39,7 @@ __device__ __forceinline__ void stage16(const char* src,
       (__attribute__((address_space(3))) void*)dst, 16, 0, 0);
 }
 
-template <typename OT, int WGM, int WGN>
+template <typename OT, int WGM, int WGN, bool ACC = false>
 __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     OT* __restrict__ C, int M, int N, int K) {
@@ -156,7 +156,9 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
       #pragma unroll
       for (int nj = 0; nj < 4; ++nj) {
         long n = tile_n + wc * 64 + nj * 16 + ccol;
-        if (n < N) C[m * N + n] = (OT)acc[mi][nj][r];
+        if (n < N)
+          C[m * N + n] = ACC ? (OT)((float)C[m * N + n] + acc[mi][nj][r])
+                             : (OT)acc[mi][nj][r];
       }
     }
   }
@@ -186,7 +188,7 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
 #define B2_TILE_BYTES (B2_BM * BK * 2)   // 16 KB per operand per K-step
 #define B2_SLOT (2 * B2_TILE_BYTES)      // A+B per ring slot
 
-template <typename OT>
+template <typename OT, bool ACC = false>
 __global__ __launch_bounds__(512, 1) void gemm_bt256_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     OT* __restrict__ C, int M, int N, int K) {
@@ -294,7 +296,9 @@ __global__ __launch_bounds__(512, 1) void gemm_bt256_kernel(
       #pragma unroll
       for (int nj = 0; nj < 4; ++nj) {
         long n = tile_n + wc * 64 + nj * 16 + ccol;
-        if (n < N) C[m * N + n] = (OT)acc[mi][nj][r];
+        if (n < N)
+          C[m * N + n] = ACC ? (OT)((float)C[m * N + n] + acc[mi][nj][r])
+                             : (OT)acc[mi][nj][r];
       }
     }
   }
@@ -316,8 +320,18 @@ __global__ void mfma_probe_kernel(const short* __restrict__ a,
 
 extern "C" {
 
+void tfosr_gemm_bt_acc(const void*, const void*, void*, int, int, int, int,
+                       int, hipStream_t);
+
 void tfosr_gemm_bt(const void* A, const void* B, void* C, int out_bf16,
                    int M, int N, int K, hipStream_t s) {
+  tfosr_gemm_bt_acc(A, B, C, out_bf16, M, N, K, 0, s);
+}
+
+// accum != 0: C += A @ B^T (the ResNet residual-join gradient accumulation
+// fused into the conv1x1 dgrad epilogue instead of a separate eager add)
+void tfosr_gemm_bt_acc(const void* A, const void* B, void* C, int out_bf16,
+                       int M, int N, int K, int accum, hipStream_t s) {
   // 256^2 4-deep-pipelined kernel for large-K tiles (917 TF @4096^3 vs 644
   // for the 2-buffer kernel; the deep prologue loses on the skinny-K conv
   // shapes, which stay on the 128^2/256x64 kernels). Toggle: TFOS_GEMM256=off
@@ -329,12 +343,20 @@ void tfosr_gemm_bt(const void* A, const void* B, void* C, int out_bf16,
   if (use256 && M >= 1024 && N >= 192 && K >= 1024) {
     int ntm = (M + B2_BM - 1) / B2_BM, ntn = (N + B2_BN - 1) / B2_BN;
     dim3 grid(ntm * ntn);
-    if (out_bf16)
-      hipLaunchKernelGGL(gemm_bt256_kernel<bf16_t>, grid, dim3(512), 0, s,
-                         (const bf16_t*)A, (const bf16_t*)B, (bf16_t*)C, M, N, K);
-    else
-      hipLaunchKernelGGL(gemm_bt256_kernel<float>, grid, dim3(512), 0, s,
-                         (const bf16_t*)A, (const bf16_t*)B, (float*)C, M, N, K);
+    if (out_bf16) {
+      if (accum)
+        hipLaunchKernelGGL((gemm_bt256_kernel<bf16_t, true>), grid, dim3(512),
+                           0, s, (const bf16_t*)A, (const bf16_t*)B,
+                           (bf16_t*)C, M, N, K);
+      else
+        hipLaunchKernelGGL((gemm_bt256_kernel<bf16_t, false>), grid, dim3(512),
+                           0, s, (const bf16_t*)A, (const bf16_t*)B,
+                           (bf16_t*)C, M, N, K);
+    } else {
+      hipLaunchKernelGGL((gemm_bt256_kernel<float, false>), grid, dim3(512), 0,
+                         s, (const bf16_t*)A, (const bf16_t*)B, (float*)C, M,
+                         N, K);
+    }
     return;
   }
   // skinny-N tile when it reduces waste (Cout=64 conv1x1 layers)
@@ -342,22 +364,37 @@ void tfosr_gemm_bt(const void* A, const void* B, void* C, int out_bf16,
   if (skinny) {
     int ntm = (M + 255) / 256, ntn = (N + 63) / 64;
     dim3 grid(ntm * ntn);
-    if (out_bf16)
-      hipLaunchKernelGGL((gemm_bt_kernel<bf16_t, 4, 1>), grid, dim3(256), 0, s,
-                         (const bf16_t*)A, (const bf16_t*)B, (bf16_t*)C, M, N, K);
-    else
+    if (out_bf16) {
+      if (accum)
+        hipLaunchKernelGGL((gemm_bt_kernel<bf16_t, 4, 1, true>), grid,
+                           dim3(256), 0, s, (const bf16_t*)A,
+                           (const bf16_t*)B, (bf16_t*)C, M, N, K);
+      else
+        hipLaunchKernelGGL((gemm_bt_kernel<bf16_t, 4, 1, false>), grid,
+                           dim3(256), 0, s, (const bf16_t*)A,
+                           (const bf16_t*)B, (bf16_t*)C, M, N, K);
+    } else {
       hipLaunchKernelGGL((gemm_bt_kernel<float, 4, 1>), grid, dim3(256), 0, s,
-                         (const bf16_t*)A, (const bf16_t*)B, (float*)C, M, N, K);
+                         (const bf16_t*)A, (const bf16_t*)B, (float*)C, M, N,
+                         K);
+    }
     return;
   }
   int ntm = (M + 127) / 128, ntn = (N + 127) / 128;
   dim3 grid(ntm * ntn);
-  if (out_bf16)
-    hipLaunchKernelGGL((gemm_bt_kernel<bf16_t, 2, 2>), grid, dim3(256), 0, s,
-                       (const bf16_t*)A, (const bf16_t*)B, (bf16_t*)C, M, N, K);
-  else
+  if (out_bf16) {
+    if (accum)
+      hipLaunchKernelGGL((gemm_bt_kernel<bf16_t, 2, 2, true>), grid, dim3(256),
+                         0, s, (const bf16_t*)A, (const bf16_t*)B, (bf16_t*)C,
+                         M, N, K);
+    else
+      hipLaunchKernelGGL((gemm_bt_kernel<bf16_t, 2, 2, false>), grid,
+                         dim3(256), 0, s, (const bf16_t*)A, (const bf16_t*)B,
+                         (bf16_t*)C, M, N, K);
+  } else {
     hipLaunchKernelGGL((gemm_bt_kernel<float, 2, 2>), grid, dim3(256), 0, s,
                        (const bf16_t*)A, (const bf16_t*)B, (float*)C, M, N, K);
+  }
 }
 
 void tfosr_mfma_probe(const short* a, const short* b, float* c, hipStream_t s) {

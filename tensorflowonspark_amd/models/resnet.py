@@ -39,6 +39,7 @@ class Bottleneck(nn.Module):
 
     def __init__(self, cin, width, stride=1, downsample=None):
         super().__init__()
+        self.stride = stride
         self.conv1 = conv1x1(cin, width)
         self.bnrelu1 = FusedBNReLU(width)
         self.conv2 = conv3x3(width, width, stride)  # v1.5: stride on the 3x3
@@ -47,8 +48,44 @@ class Bottleneck(nn.Module):
         # block tail bn3 + residual-add + relu as ONE fused op
         self.bn3 = FusedBNAddReLU(width * self.expansion)
         self.downsample = downsample
+        # whole-block fused path (manual backward: residual-join grad adds
+        # fused into the conv dgrad epilogues): requires every conv on the
+        # MFMA kernels and a (Conv1x1, FusedBN) downsample
+        from ..ops.modules import Conv1x1, Conv3x3, FusedBN
+        self._block_fusable = (
+            isinstance(self.conv1, Conv1x1) and isinstance(self.conv2, Conv3x3)
+            and isinstance(self.conv3, Conv1x1) and cin % 32 == 0
+            and (downsample is None or
+                 (isinstance(downsample[0], Conv1x1)
+                  and type(downsample[1]) is FusedBN)))
 
     def forward(self, x):
+        import os
+        if (self.training and x.is_cuda and x.dtype == torch.bfloat16
+                and self._block_fusable
+                and os.environ.get("TFOS_FUSED_BLOCK", "on") != "off"):
+            from ..ops import get_ext
+            if get_ext(required=True) is not None:
+                from ..ops.modules import _BottleneckFn
+                x = x.contiguous(memory_format=torch.channels_last)
+                bn1, bn2, bn3 = self.bnrelu1, self.bnrelu2, self.bn3
+                for bn in (bn1, bn2, bn3):
+                    bn.num_batches_tracked += 1
+                if self.downsample is not None:
+                    dconv, dbn = self.downsample[0], self.downsample[1]
+                    dbn.num_batches_tracked += 1
+                    dargs = (dconv.weight, dbn.weight, dbn.bias,
+                             dbn.running_mean, dbn.running_var)
+                else:
+                    dargs = (None, None, None, None, None)
+                return _BottleneckFn.apply(
+                    x, self.conv1.weight, bn1.weight, bn1.bias,
+                    bn1.running_mean, bn1.running_var,
+                    self.conv2.weight, bn2.weight, bn2.bias,
+                    bn2.running_mean, bn2.running_var,
+                    self.conv3.weight, bn3.weight, bn3.bias,
+                    bn3.running_mean, bn3.running_var,
+                    *dargs, self.stride, bn1.momentum, bn1.eps)
         identity = x
         out = self.bnrelu1(self.conv1(x))
         out = self.bnrelu2(self.conv2(out))

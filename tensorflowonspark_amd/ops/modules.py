@@ -743,3 +743,158 @@ class BucketAdam:
 
     def zero_grad(self):
         self.engine.zero_grad()
+
+
+# ---------------------------------------------------------------------------
+# Fully-fused ResNet Bottleneck (manual backward, zero autograd glue)
+# ---------------------------------------------------------------------------
+
+class _BottleneckFn(torch.autograd.Function):
+    """One autograd node for the whole bottleneck block.
+
+    Eager autograd accumulates the residual-join gradients (the block input
+    feeds both conv1 and the identity/downsample path) with whole-tensor adds
+    — 8.4 ms/step of CUDAFunctor_add at ResNet-50 b1024 (profiles/README
+    r01). The manual backward instead makes the conv1 (and downsample)
+    backward-data kernels accumulate straight into the residual gradient
+    buffer (gemm_bt_acc / conv_mfma_acc epilogues), so no separate add runs.
+
+    Tensors bf16 channels_last; BN stats/params fp32. Training mode only;
+    saved set (t*, a*, masks, stats) matches what the unfused composition
+    saves, so peak memory is unchanged.
+    """
+
+    @staticmethod
+    def forward(ctx, x, w1, g1, b1, rm1, rv1, w2, g2, b2, rm2, rv2,
+                w3, g3, b3, rm3, rv3, wd, gd, bd, rmd, rvd,
+                stride, momentum, eps):
+        ext = get_ext(required=True)
+        N, Cin, H, W = x.shape
+        C1, C2, C3 = w1.shape[0], w2.shape[0], w3.shape[0]
+
+        def as2d(t):
+            return t.permute(0, 2, 3, 1).reshape(-1, t.shape[1])
+
+        def as4d(t2, h, w):
+            return t2.view(N, h, w, -1).permute(0, 3, 1, 2)
+
+        w1b = w1.view(C1, Cin).to(torch.bfloat16).contiguous()
+        t1 = as4d(ext.gemm_bt(as2d(x), w1b, True), H, W)
+        a1, m1, r1, k1 = ext.bn_fwd_train(t1, None, g1, b1, rm1, rv1,
+                                          momentum, eps, True)
+        w9 = w2.permute(0, 2, 3, 1).reshape(C2, 9 * C1) \
+            .to(torch.bfloat16).contiguous()
+        t2 = ext.conv_mfma(a1, w9, C2, 3, 3, stride, 1, 1, -1, -1)
+        a2, m2, r2, k2 = ext.bn_fwd_train(t2, None, g2, b2, rm2, rv2,
+                                          momentum, eps, True)
+        OH, OW = t2.shape[2], t2.shape[3]
+        w3b = w3.view(C3, C2).to(torch.bfloat16).contiguous()
+        t3 = as4d(ext.gemm_bt(as2d(a2), w3b, True), OH, OW)
+
+        if wd is not None:
+            wdb = wd.view(C3, Cin).to(torch.bfloat16).contiguous()
+            if stride == 1:
+                td = as4d(ext.gemm_bt(as2d(x), wdb, True), H, W)
+            else:
+                td = ext.conv_mfma(x, wdb, C3, 1, 1, stride, 0, 1, -1, -1)
+            idn, md, rd, _kd = ext.bn_fwd_train(td, None, gd, bd, rmd, rvd,
+                                                momentum, eps, False)
+            opt = [td, idn, md, rd]
+        else:
+            idn = None
+            opt = []
+        res = idn if idn is not None else x
+        y, m3, r3, k3 = ext.bn_fwd_train(t3, res, g3, b3, rm3, rv3,
+                                         momentum, eps, True)
+
+        ctx.save_for_backward(x, w1, g1, t1, a1, m1, r1, k1,
+                              w2, g2, t2, a2, m2, r2, k2,
+                              w3, g3, t3, m3, r3, k3, y, wd, gd, *opt)
+        ctx.meta = (stride, wd is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_ext(required=True)
+        (x, w1, g1, t1, a1, m1, r1, k1, w2, g2, t2, a2, m2, r2, k2,
+         w3, g3, t3, m3, r3, k3, y, wd, gd, *opt) = ctx.saved_tensors
+        stride, has_down = ctx.meta
+        N, Cin, H, W = x.shape
+        C1, C2, C3 = w1.shape[0], w2.shape[0], w3.shape[0]
+        OH, OW = t2.shape[2], t2.shape[3]
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        if dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
+
+        def as2d(t):
+            return t.permute(0, 2, 3, 1).reshape(-1, t.shape[1])
+
+        def as4d(t2d, h, w):
+            return t2d.view(N, h, w, -1).permute(0, 3, 1, 2)
+
+        def cl(t):
+            return t.contiguous(memory_format=torch.channels_last)
+
+        # block tail: bn3(+res+relu) backward -> dt3 plus the gated residual
+        # gradient dres (doubles as the dx accumulation buffer below)
+        dt3, dg3, db3, dres = ext.bn_bwd(t3, dy, y, k3, g3, m3, r3,
+                                         True, True)
+        dt3 = cl(dt3)
+
+        # conv3 (1x1): dgrad + wrw
+        w3b = w3.view(C3, C2).to(torch.bfloat16)
+        da2 = cl(as4d(ext.gemm_bt(as2d(dt3), w3b.t().contiguous(), True),
+                      OH, OW))
+        dw3 = ext.conv_wrw2(dt3, a2, 1, 1, 1, 0).view(C3, C2, 1, 1)
+
+        # bn2+relu backward
+        dt2, dg2, db2 = ext.bn_bwd(t2, da2, a2, k2, g2, m2, r2, True, False)
+        dt2 = cl(dt2)
+
+        # conv2 (3x3, stride s): dgrad (input-dilated) + wrw
+        w9p = w2.flip(2, 3).permute(1, 2, 3, 0).reshape(C1, 9 * C2) \
+            .to(torch.bfloat16).contiguous()
+        da1 = ext.conv_mfma(dt2, w9p, C1, 3, 3, 1, 1, stride, H, W)
+        dw9 = ext.conv_wrw2(dt2, a1, 3, 3, stride, 1)
+        dw2 = dw9.view(C2, 3, 3, C1).permute(0, 3, 1, 2).contiguous()
+
+        # bn1+relu backward
+        dt1, dg1, db1 = ext.bn_bwd(t1, da1, a1, k1, g1, m1, r1, True, False)
+        dt1 = cl(dt1)
+
+        # conv1 (1x1) wrw
+        dw1 = ext.conv_wrw2(dt1, x, 1, 1, 1, 0).view(C1, Cin, 1, 1)
+
+        w1b = w1.view(C1, Cin).to(torch.bfloat16)
+        if has_down:
+            td, idn, md, rd = opt
+            # downsample path: bnd backward (no relu) then conv dgrad
+            dtd, dgd, dbd = ext.bn_bwd(td, dres, idn,
+                                       torch.empty(0, dtype=torch.uint8,
+                                                   device=td.device),
+                                       gd, md, rd, False, False)
+            dtd = cl(dtd)
+            dwd = ext.conv_wrw2(dtd, x, 1, 1, stride, 0) \
+                .view(C3, Cin, 1, 1)
+            # dx = conv1_dgrad, then downsample dgrad ACCUMULATES into it
+            dx2d = ext.gemm_bt(as2d(dt1), w1b.t().contiguous(), True)
+            dx = cl(as4d(dx2d, H, W))
+            wdb = wd.view(C3, Cin).to(torch.bfloat16)
+            if stride == 1:
+                ext.gemm_bt_acc(as2d(dtd), wdb.t().contiguous(), as2d(dx))
+            else:
+                ext.conv_mfma_acc(dtd, wdb.t().contiguous(), dx,
+                                  1, 1, 1, 0, stride)
+            dwd = dwd.to(wd.dtype)
+            dgd_, dbd_ = dgd, dbd
+        else:
+            # dx = dres + conv1_dgrad — accumulate straight into dres
+            dx = dres
+            ext.gemm_bt_acc(as2d(dt1), w1b.t().contiguous(), as2d(dx))
+            dwd = dgd_ = dbd_ = None
+
+        return (dx, dw1.to(w1.dtype), dg1, db1, None, None,
+                dw2.to(w2.dtype), dg2, db2, None, None,
+                dw3.to(w3.dtype), dg3, db3, None, None,
+                dwd, dgd_, dbd_, None, None,
+                None, None, None)

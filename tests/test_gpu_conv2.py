@@ -194,3 +194,49 @@ def test_conv_wrw2_kernel(case):
     y.backward(dy.float().cpu())
     ref = w_ref.grad.permute(0, 2, 3, 1).reshape(Cout, k * k * Cin)
     _close(dw, ref, 2e-2, 2e-1 * max(1, N // 2), "wrw2")
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("cfg", [
+    (64, 64, 1, True),     # layer1 block1: downsample, stride 1
+    (256, 64, 1, False),   # layer1 later blocks: identity
+    (256, 128, 2, True),   # layer2 block1: downsample, stride 2
+    (512, 128, 1, False),
+])
+def test_fused_bottleneck_matches_eager(cfg):
+    """Whole-block fused backward (residual-join adds fused into conv dgrad
+    epilogues) vs the eager composition."""
+    import os
+    from tensorflowonspark_amd.models.resnet import Bottleneck, conv1x1
+    from tensorflowonspark_amd.ops.modules import FusedBN
+    import torch.nn as nn
+    torch.manual_seed(11)
+    cin, width, stride, down = cfg
+    ds = None
+    if down:
+        ds = nn.Sequential(conv1x1(cin, width * 4, stride),
+                           FusedBN(width * 4))
+    blk = Bottleneck(cin, width, stride, ds).cuda() \
+        .to(memory_format=torch.channels_last)
+    blk.train()
+    assert blk._block_fusable
+    x = torch.randn(4, cin, 16, 16, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+
+    def run(env):
+        os.environ["TFOS_FUSED_BLOCK"] = env
+        blk.zero_grad(set_to_none=True)
+        xg = x.clone().requires_grad_(True)
+        y = blk(xg)
+        y.float().square().mean().backward()
+        grads = {n: p.grad.float().clone() for n, p in blk.named_parameters()}
+        return y.float().clone(), xg.grad.float().clone(), grads
+
+    y_f, dx_f, g_f = run("on")
+    y_e, dx_e, g_e = run("off")
+    _close(y_f, y_e, 2e-2, 1e-1, "fused fwd")
+    _close(dx_f, dx_e, 3e-2, 5e-2, "fused dx")
+    for n in g_e:
+        _close(g_f[n], g_e[n], 3e-2, 2e-1, "grad " + n)
+    os.environ.pop("TFOS_FUSED_BLOCK", None)
