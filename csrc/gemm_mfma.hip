@@ -23,6 +23,7 @@
 // blockIdx -> tile mapping is XCD-aware (8 XCDs with private L2 on MI355X):
 // contiguous grid chunks land on one XCD so neighboring tiles share L2.
 #include "tfosr_common.h"
+#include <type_traits>
 #include <cstdlib>
 #include <cstring>
 
@@ -30,6 +31,7 @@ typedef short bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define BK 32
+#define AS_LDS(T) __attribute__((address_space(3))) T
 #define LDS_SWZ(l) ((l) ^ ((((l) >> 7) & 7) << 4))
 
 __device__ __forceinline__ void stage16(const char* src,
@@ -147,6 +149,63 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
   // epilogue: C/D layout col = lane&15, row = (lane>>4)*4 + reg
   const int ccol = lane & 15;
   const int crow0 = (lane >> 4) * 4;
+  if constexpr (ACC && std::is_same<OT, bf16_t>::value) {
+    // shuffle epilogue: the mfma lane layout reads/writes C in scattered
+    // 32 B runs — fine for pure stores (write combining) but a read-modify-
+    // write at that pattern costs as much as a separate eager add. Stage the
+    // tile through LDS (8 KB bf16 per wave, reusing the staging ring) and do
+    // the += with fully-coalesced 16 B-per-lane loads/stores.
+    __syncthreads();  // staging LDS ring is dead now
+    AS_LDS(char*) myl = lds3 + wave * 8192;
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+      #pragma unroll
+      for (int r = 0; r < 4; ++r)
+        #pragma unroll
+        for (int nj = 0; nj < 4; ++nj) {
+          bf16_t v = (bf16_t)acc[mi][nj][r];
+          short sv;
+          __builtin_memcpy(&sv, &v, 2);
+          *(AS_LDS(short*))(myl +
+              ((mi * 16 + crow0 + r) * 64 + nj * 16 + ccol) * 2) = sv;
+        }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    typedef short s16x8 __attribute__((ext_vector_type(8)));
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      int c = i * 64 + lane;          // 512 16B-chunks: row = c>>3, col8 = c&7
+      int row = c >> 3, col0 = (c & 7) * 8;
+      long m = tile_m + wr * 64 + row;
+      long n = tile_n + wc * 64 + col0;
+      if (m >= M || n + 7 >= N) {     // slow scalar tail at ragged edges
+        if (m < M)
+          for (int j = 0; j < 8 && n + j < N; ++j) {
+            short sv = *(AS_LDS(short*))(myl + (row * 64 + col0 + j) * 2);
+            bf16_t v;
+            __builtin_memcpy(&v, &sv, 2);
+            C[m * N + n + j] =
+                (OT)((float)C[m * N + n + j] + (float)v);
+          }
+        continue;
+      }
+      s16x8 add = *(AS_LDS(s16x8*))(myl + (row * 64 + col0) * 2);
+      s16x8 old = *(s16x8*)((char*)C + (m * N + n) * 2);
+      s16x8 out;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        short sa = add[j], sb = old[j];
+        bf16_t a, b;
+        __builtin_memcpy(&a, &sa, 2);
+        __builtin_memcpy(&b, &sb, 2);
+        bf16_t o = (bf16_t)((float)a + (float)b);
+        short so;
+        __builtin_memcpy(&so, &o, 2);
+        out[j] = so;
+      }
+      *(s16x8*)((char*)C + (m * N + n) * 2) = out;
+    }
+    return;
+  }
   #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
     #pragma unroll
