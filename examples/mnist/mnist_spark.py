@@ -37,6 +37,12 @@ def map_fun(args, ctx):
     opt = BucketSGD(engine, lr=args.lr, momentum=0.9)
     feed = TFNode.DataFeed(ctx.mgr, train_mode=True)
     timer = StepTimer(args.batch_size, log_every=50)
+    # chief writes TensorBoard scalar events so the TFCluster-spawned
+    # TensorBoard (tensorboard_url) serves populated curves
+    writer = None
+    if ctx.is_chief and getattr(args, "model_dir", None):
+        from tensorflowonspark_amd.utils.events import SummaryWriter
+        writer = SummaryWriter(args.model_dir)
     model.train()
 
     step = start_step
@@ -57,12 +63,17 @@ def map_fun(args, ctx):
         opt.step()
         timer.step()
         step += 1
+        if writer is not None and step % 10 == 0:
+            writer.add_scalars({"loss": float(loss),
+                                "images_per_sec": timer.rate()}, step)
         if step % 200 == 0 and ctx.is_chief:
             ckpt.save_checkpoint(args.model_dir, step, model)
 
     # drain anything this rank didn't consume (reference mnist_spark.py:71 —
     # leftover queued blocks would otherwise trip the feeder's feed_timeout)
     feed.terminate()
+    if writer is not None:
+        writer.close()
     if ctx.is_chief:
         ckpt.save_checkpoint(args.model_dir, step, model)
         ctx.export_saved_model(model.cpu(), args.export_dir)

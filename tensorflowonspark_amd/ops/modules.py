@@ -667,6 +667,16 @@ class DenseMFMA(nn.Module):
         return F.linear(x, self.weight.to(x.dtype),
                         None if self.bias is None else self.bias.to(x.dtype))
 
+    def __prepare_scriptable__(self):
+        # TorchScript export swaps in the equivalent nn.Linear (the custom
+        # kernel dispatch isn't scriptable; serving runs the library op)
+        m = nn.Linear(self.weight.shape[1], self.weight.shape[0],
+                      bias=self.bias is not None)
+        m.weight = self.weight
+        if self.bias is not None:
+            m.bias = self.bias
+        return m
+
 
 class Conv2dIm2colMFMA(nn.Module):
     """Small-Cin conv (e.g. MNIST's Conv2D(1->32, k3)) via im2col into the
@@ -700,6 +710,14 @@ class Conv2dIm2colMFMA(nn.Module):
             return y2.view(N, OH * OW, Cout).transpose(1, 2)                 .reshape(N, Cout, OH, OW)
         b = None if self.bias is None else self.bias.to(x.dtype)
         return F.conv2d(x, self.weight.to(x.dtype), b)
+
+    def __prepare_scriptable__(self):
+        m = nn.Conv2d(self.weight.shape[1], self.weight.shape[0], self.k,
+                      bias=self.bias is not None)
+        m.weight = self.weight
+        if self.bias is not None:
+            m.bias = self.bias
+        return m
 
 
 class BucketAdam:
@@ -841,11 +859,23 @@ class _BottleneckFn(torch.autograd.Function):
                                          True, True)
         dt3 = cl(dt3)
 
+        import os
+        wrw2 = os.environ.get("TFOS_WRW", "mfma2") == "mfma2"
+
+        def lib_wrw(dyt, xt, w4, s_, p_):
+            w4 = w4.to(torch.bfloat16).contiguous(
+                memory_format=torch.channels_last)
+            _, dw, _ = torch.ops.aten.convolution_backward(
+                dyt, xt, w4, None, [s_, s_], [p_, p_], [1, 1], False, [0, 0],
+                1, [False, True, False])
+            return dw
+
         # conv3 (1x1): dgrad + wrw
         w3b = w3.view(C3, C2).to(torch.bfloat16)
         da2 = cl(as4d(ext.gemm_bt(as2d(dt3), w3b.t().contiguous(), True),
                       OH, OW))
-        dw3 = ext.conv_wrw2(dt3, a2, 1, 1, 1, 0).view(C3, C2, 1, 1)
+        dw3 = ext.conv_wrw2(dt3, a2, 1, 1, 1, 0).view(C3, C2, 1, 1) if wrw2 \
+            else lib_wrw(dt3, a2, w3.view(C3, C2, 1, 1), 1, 0)
 
         # bn2+relu backward
         dt2, dg2, db2 = ext.bn_bwd(t2, da2, a2, k2, g2, m2, r2, True, False)
@@ -855,15 +885,19 @@ class _BottleneckFn(torch.autograd.Function):
         w9p = w2.flip(2, 3).permute(1, 2, 3, 0).reshape(C1, 9 * C2) \
             .to(torch.bfloat16).contiguous()
         da1 = ext.conv_mfma(dt2, w9p, C1, 3, 3, 1, 1, stride, H, W)
-        dw9 = ext.conv_wrw2(dt2, a1, 3, 3, stride, 1)
-        dw2 = dw9.view(C2, 3, 3, C1).permute(0, 3, 1, 2).contiguous()
+        if wrw2:
+            dw9 = ext.conv_wrw2(dt2, a1, 3, 3, stride, 1)
+            dw2 = dw9.view(C2, 3, 3, C1).permute(0, 3, 1, 2).contiguous()
+        else:
+            dw2 = lib_wrw(dt2, a1, w2, stride, 1)
 
         # bn1+relu backward
         dt1, dg1, db1 = ext.bn_bwd(t1, da1, a1, k1, g1, m1, r1, True, False)
         dt1 = cl(dt1)
 
         # conv1 (1x1) wrw
-        dw1 = ext.conv_wrw2(dt1, x, 1, 1, 1, 0).view(C1, Cin, 1, 1)
+        dw1 = ext.conv_wrw2(dt1, x, 1, 1, 1, 0).view(C1, Cin, 1, 1) if wrw2 \
+            else lib_wrw(dt1, x, w1.view(C1, Cin, 1, 1), 1, 0)
 
         w1b = w1.view(C1, Cin).to(torch.bfloat16)
         if has_down:
@@ -875,7 +909,8 @@ class _BottleneckFn(torch.autograd.Function):
                                        gd, md, rd, False, False)
             dtd = cl(dtd)
             dwd = ext.conv_wrw2(dtd, x, 1, 1, stride, 0) \
-                .view(C3, Cin, 1, 1)
+                .view(C3, Cin, 1, 1) if wrw2 \
+                else lib_wrw(dtd, x, wd.view(C3, Cin, 1, 1), stride, 0)
             # dx = conv1_dgrad, then downsample dgrad ACCUMULATES into it
             dx2d = ext.gemm_bt(as2d(dt1), w1b.t().contiguous(), True)
             dx = cl(as4d(dx2d, H, W))

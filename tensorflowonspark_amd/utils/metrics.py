@@ -53,8 +53,13 @@ class StepTimer:
                         self.name, self.total_steps, ips, ms)
             self._t0 = time.time()
             self._window_steps = 0
+            self._last_rate = ips
             return ips
         return None
+
+    def rate(self):
+        """Most recent windowed images/sec (0.0 before the first window)."""
+        return getattr(self, "_last_rate", 0.0)
 
 
 @contextlib.contextmanager

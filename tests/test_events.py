@@ -1,0 +1,34 @@
+"""TensorBoard event-file writer: records round-trip
+through our own codec and follow the TFRecord/Event wire format."""
+
+import glob
+
+from tensorflowonspark_amd import tfrecord
+from tensorflowonspark_amd.utils import events
+
+
+def test_scalar_event_roundtrip():
+    rec = events.encode_scalar_event(42, {"loss": 1.5, "acc": 0.75},
+                                     wall_time=123.25)
+    wall, step, scalars = events.decode_scalar_event(rec)
+    assert wall == 123.25 and step == 42
+    assert abs(scalars["loss"] - 1.5) < 1e-6
+    assert abs(scalars["acc"] - 0.75) < 1e-6
+
+
+def test_summary_writer_file(tmp_path):
+    logdir = str(tmp_path / "run1")
+    with events.SummaryWriter(logdir) as w:
+        for s in range(5):
+            w.add_scalar("loss", 2.0 / (s + 1), s)
+        w.add_scalars({"imgs_per_sec": 9000.0, "lr": 0.1}, 4)
+    files = glob.glob(logdir + "/events.out.tfevents.*")
+    assert len(files) == 1
+    recs = list(tfrecord.tfrecord_iterator(files[0], verify=True))
+    assert len(recs) == 7  # header + 5 + 1
+    # header carries file_version
+    assert b"brain.Event:2" in recs[0]
+    wall, step, sc = events.decode_scalar_event(recs[3])
+    assert step == 2 and abs(sc["loss"] - 2.0 / 3) < 1e-6
+    _, step6, sc6 = events.decode_scalar_event(recs[6])
+    assert step6 == 4 and abs(sc6["imgs_per_sec"] - 9000.0) < 1e-3
