@@ -63,6 +63,9 @@ class DDPEngine:
 
         self._comm_stream = (torch.cuda.Stream()
                              if torch.cuda.is_available() else None)
+        #: optional callback(bucket) fired when a bucket's grads are all
+        #: accumulated (used by AsyncSGD to overlap PS push with backward)
+        self.bucket_ready_cb = None
         self._buckets = self._build_buckets(params)
         self._hooks = []
         for bucket in self._buckets:
@@ -121,6 +124,8 @@ class DDPEngine:
                 return
             bucket.ready += 1
             if bucket.ready == len(bucket.params):
+                if self.bucket_ready_cb is not None:
+                    self.bucket_ready_cb(bucket)
                 self._reduce_bucket(bucket)
         return hook
 

@@ -26,6 +26,19 @@ import numpy as np
 logger = logging.getLogger(__name__)
 
 
+def _f32_to_bf16(arr_f32):
+    """fp32 ndarray -> raw bf16 bytes (round-to-nearest-even)."""
+    u = arr_f32.view(np.uint32)
+    r = (u + 0x7FFF + ((u >> 16) & 1)) >> 16
+    return r.astype(np.uint16).tobytes()
+
+
+def _bf16_to_f32(raw):
+    """raw bf16 bytes -> fp32 ndarray."""
+    u = np.frombuffer(raw, dtype=np.uint16).astype(np.uint32) << 16
+    return u.view(np.float32)
+
+
 def _send_msg(sock, header, payload=b""):
     """header: small picklable tuple (json-free for bytes); payload: raw."""
     import pickle
@@ -63,7 +76,10 @@ class ParameterServer:
     Protocol (header tuple, payload raw bytes):
       ("init", bucket_id, nelem)  + fp32 payload -> ack (first writer wins)
       ("pull", bucket_id)                        -> fp32 payload
-      ("push", bucket_id, lr, momentum, wd)      + fp32 grad -> fp32 params
+      ("push", bucket_id, lr, momentum, wd[, dtype]) + grad -> params
+          dtype "f32" (default) or "bf16": wire format of BOTH the pushed
+          gradient and the returned parameters (halves the TCP bytes per
+          step; master params stay fp32 server-side)
       ("stop",)                                  -> ack, server exits
     """
 
@@ -122,8 +138,12 @@ class ParameterServer:
                         data = self._params[bid].tobytes()
                     _send_msg(conn, ("ok",), data)
                 elif op == "push":
-                    _, bid, lr, mu, wd = header
-                    grad = np.frombuffer(payload, dtype=np.float32)
+                    bid, lr, mu, wd = header[1:5]
+                    dtype = header[5] if len(header) > 5 else "f32"
+                    if dtype == "bf16":
+                        grad = _bf16_to_f32(payload)
+                    else:
+                        grad = np.frombuffer(payload, dtype=np.float32)
                     with self._lock:
                         p = self._params[bid]
                         m = self._mom[bid]
@@ -131,7 +151,8 @@ class ParameterServer:
                         np.multiply(m, mu, out=m)
                         np.add(m, g, out=m)
                         p -= lr * m
-                        data = p.tobytes()
+                        data = _f32_to_bf16(p) if dtype == "bf16" \
+                            else p.tobytes()
                     _send_msg(conn, ("ok",), data)
                 elif op == "stop":
                     _send_msg(conn, ("ok",))
@@ -185,12 +206,23 @@ class PSClient:
         _, payload = _recv_msg(s)
         out_np[:] = np.frombuffer(payload, dtype=np.float32)
 
-    def push_pull(self, bid, grad_np, out_np, lr, momentum=0.9, weight_decay=0.0):
-        """Push a gradient, receive the updated parameters (one round trip)."""
+    def push_pull(self, bid, grad_np, out_np, lr, momentum=0.9,
+                  weight_decay=0.0, wire="f32"):
+        """Push a gradient, receive the updated parameters (one round trip).
+
+        ``wire="bf16"`` halves the bytes both ways (grad pushed and params
+        returned in bf16; the server's master copy stays fp32)."""
         s = self._sock(self._shard(bid))
-        _send_msg(s, ("push", bid, lr, momentum, weight_decay), grad_np.tobytes())
-        _, payload = _recv_msg(s)
-        out_np[:] = np.frombuffer(payload, dtype=np.float32)
+        if wire == "bf16":
+            _send_msg(s, ("push", bid, lr, momentum, weight_decay, "bf16"),
+                      _f32_to_bf16(grad_np))
+            _, payload = _recv_msg(s)
+            out_np[:] = _bf16_to_f32(payload)
+        else:
+            _send_msg(s, ("push", bid, lr, momentum, weight_decay),
+                      grad_np.tobytes())
+            _, payload = _recv_msg(s)
+            out_np[:] = np.frombuffer(payload, dtype=np.float32)
 
     def stop_all(self):
         for shard in range(len(self.addrs)):
@@ -214,17 +246,33 @@ class AsyncSGD:
 
     Use with ``DDPEngine(model, broadcast_params=False, flatten_params=True)``
     and *no* process group (each worker runs independently; consistency comes
-    from the server). ``step()`` pushes each bucket's gradient and installs the
-    returned parameters — overlap is per-bucket (push bucket i while i+1's
-    grad is still being copied off-device).
+    from the server).
+
+    ``overlap=True`` (default): each bucket's push/pull round trip starts from
+    the grad-ready hook DURING backward — the TCP transfer of early (output-
+    side) buckets runs while the GPU is still computing input-side gradients.
+    ``step()`` only joins the in-flight round trips and installs the returned
+    parameters. Round 1 did a serial post-step loop over all buckets
+    (~100 MB/step blocking the GPU). ``wire="bf16"``
+    additionally halves the bytes on the wire.
     """
 
-    def __init__(self, engine, client, lr=0.01, momentum=0.9, weight_decay=0.0):
+    def __init__(self, engine, client, lr=0.01, momentum=0.9,
+                 weight_decay=0.0, wire="bf16", overlap=True):
+        import torch
         self.engine = engine
         self.client = client
         self.lr = lr
         self.momentum = momentum
         self.weight_decay = weight_decay
+        self.wire = wire
+        self.overlap = overlap
+        self._cuda = torch.cuda.is_available() and \
+            engine._buckets and engine._buckets[0].buffer.is_cuda
+        self._threads = {}
+        self._host_out = {}
+        self._pin = {}
+        self._lock = threading.Lock()
         for i, bucket in enumerate(engine._buckets):
             pf = bucket.param_flat
             assert pf is not None, "AsyncSGD requires flatten_params=True"
@@ -233,6 +281,43 @@ class AsyncSGD:
             host = np.empty(pf.numel(), dtype=np.float32)
             client.pull(i, host)
             with_torch_copy(pf, host)
+            self._host_out[i] = np.empty(pf.numel(), dtype=np.float32)
+            self._pin[i] = torch.empty(bucket.buffer.numel(),
+                                       dtype=torch.float32,
+                                       pin_memory=self._cuda)
+        if overlap:
+            self._bucket_idx = {id(b): i
+                                for i, b in enumerate(engine._buckets)}
+            engine.bucket_ready_cb = self._on_bucket_ready
+
+    def _on_bucket_ready(self, bucket):
+        """Grad-ready hook (fires inside backward): stage the bucket's grad
+        off-device asynchronously and start its PS round trip on a thread."""
+        import torch
+        i = self._bucket_idx[id(bucket)]
+        ev = None
+        if self._cuda:
+            stream = self.engine._comm_stream or torch.cuda.current_stream()
+            if self.engine._comm_stream is not None:
+                stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                self._pin[i].copy_(bucket.buffer.float(), non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record(stream)
+        else:
+            self._pin[i].copy_(bucket.buffer.float())
+        t = threading.Thread(target=self._roundtrip, args=(i, ev), daemon=True)
+        t.start()
+        self._threads[i] = t
+
+    def _roundtrip(self, i, ev):
+        if ev is not None:
+            ev.synchronize()
+        grad = self._pin[i].numpy()
+        with self._lock:  # one TCP conversation per socket at a time
+            self.client.push_pull(i, grad, self._host_out[i], self.lr,
+                                  self.momentum, self.weight_decay,
+                                  wire=self.wire)
 
     @staticmethod
     def _np(t):
@@ -242,11 +327,18 @@ class AsyncSGD:
         self.engine.zero_grad()
 
     def step(self):
+        if self.overlap and self._threads:
+            for i, t in sorted(self._threads.items()):
+                t.join()
+                with_torch_copy(self.engine._buckets[i].param_flat,
+                                self._host_out[i])
+            self._threads = {}
+            return
         for i, bucket in enumerate(self.engine._buckets):
             grad = self._np(bucket.buffer)
-            out = np.empty_like(grad)
+            out = self._host_out[i]
             self.client.push_pull(i, grad, out, self.lr, self.momentum,
-                                  self.weight_decay)
+                                  self.weight_decay, wire=self.wire)
             with_torch_copy(bucket.param_flat, out)
 
 

@@ -90,3 +90,48 @@ def test_async_ps_training():
             assert last < first, "async training did not reduce loss"
     finally:
         sc.stop()
+
+
+def test_async_sgd_overlap_and_bf16_wire():
+    """push/pull round trips start from the grad-ready
+    hooks during backward (not a serial post-step loop), and the bf16 wire
+    format round-trips through the server."""
+    import threading
+
+    import numpy as np
+    import torch
+
+    from tensorflowonspark_amd.parallel import DDPEngine
+    from tensorflowonspark_amd.parallel.ps import (AsyncSGD, ParameterServer,
+                                                   PSClient, _bf16_to_f32,
+                                                   _f32_to_bf16)
+
+    # wire converters round trip within bf16 precision
+    a = np.linspace(-3, 3, 64, dtype=np.float32)
+    b = _bf16_to_f32(_f32_to_bf16(a))
+    assert np.abs(a - b).max() < 0.02
+
+    srv = ParameterServer()
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    client = PSClient(["127.0.0.1:{}".format(srv.port)])
+    model = torch.nn.Linear(1, 1, bias=False)
+    eng = DDPEngine(model, bucket_mb=1, broadcast_params=False)
+    opt = AsyncSGD(eng, client, lr=0.2, momentum=0.0, wire="bf16")
+    assert eng.bucket_ready_cb is not None  # overlap installed
+
+    for _ in range(150):
+        x = torch.rand(32, 1)
+        y = 2 * x
+        opt.zero_grad()
+        loss = torch.nn.functional.mse_loss(model(x), y)
+        loss.backward()
+        # round trips already in flight before step() (overlap)
+        in_flight = len(opt._threads)
+        eng.finalize_backward()
+        opt.step()
+        assert in_flight >= 1
+    w = float(model.weight.flatten()[0])
+    assert abs(w - 2.0) < 0.1, w
+    client.stop_all()
+    srv.stop()
