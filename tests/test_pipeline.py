@@ -146,3 +146,22 @@ def test_transform_with_input_shapes(tmp_path):
         assert len(preds[0][0]) == 10  # 10 logits per row
     finally:
         sc.stop()
+
+
+def test_params_are_pyspark_params_when_available():
+    """With pyspark installed, TFEstimator/TFModel must be real
+    pyspark.ml.param.Params (and Pipeline stages); skipped otherwise."""
+    import pytest
+    pytest.importorskip("pyspark")
+    from pyspark.ml import Estimator
+    from pyspark.ml.param import Params
+
+    from tensorflowonspark_amd.pipeline import TFEstimator
+
+    est = TFEstimator(lambda a, c: None, {})
+    assert isinstance(est, Params)
+    assert isinstance(est, Estimator)
+    est.setBatchSize(64)
+    assert est.getBatchSize() == 64
+    args = est.merge_args_params()
+    assert args.batch_size == 64
