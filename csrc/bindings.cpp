@@ -53,6 +53,8 @@ int tfosr_wrw2_split(int, int, long);
 void tfosr_conv_wrw2(const void*, const void*, const void*, float*, float*,
                      int, int, int, int, int, int, int, int, int, int, int,
                      int, hipStream_t);
+void tfosr_conv_stem(const void*, const void*, const void*, void*, int, int,
+                     int, int, int, int, hipStream_t);
 }
 
 namespace tfosr {
@@ -443,6 +445,48 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     tfosr_conv_wrw2(dy.data_ptr(), x.data_ptr(), guard.data_ptr(),
                     ws.data_ptr<float>(), dW.data_ptr<float>(), N, H, W, Cin,
                     Cout, OH, OW, R, S_f, stride, P, split, cur_stream());
+    return dW;
+  });
+  // ResNet stem: 7x7/s2/p3 conv over a pre-padded NHWC4 image
+  // ([N,4,230,230] channels_last bf16, Cin 3->4 zero-padded, spatial pad 3
+  // baked in); weight pre-packed to [Cout, 7*32] with zero columns for the
+  // pads. Output [N, Cout, 112, 112].
+  m.def("conv_stem", [](at::Tensor x4, at::Tensor w224, long Cout) {
+    TORCH_CHECK(x4.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                x4.size(1) == 4, "conv_stem expects NHWC4 padded input");
+    TORCH_CHECK(x4.scalar_type() == at::kBFloat16 &&
+                w224.scalar_type() == at::kBFloat16);
+    TORCH_CHECK(w224.size(1) == 224);
+    int N = x4.size(0), Hp = x4.size(2), Wp = x4.size(3);
+    int OH = (Hp - 7) / 2 + 1, OW = (Wp - 7) / 2 + 1;
+    auto y = at::empty({N, Cout, OH, OW}, x4.options(),
+                       at::MemoryFormat::ChannelsLast);
+    auto guard = at::zeros({64}, x4.options());
+    tfosr_conv_stem(x4.data_ptr(), w224.contiguous().data_ptr(),
+                    guard.data_ptr(), y.data_ptr(), N, Hp, Wp, Cout, OH, OW,
+                    cur_stream());
+    return y;
+  });
+  // stem weight gradient over the same NHWC4 padded view: returns fp32
+  // [Cout, 7*32] (rows x (8 px x 4 ch), pads included — caller unpacks)
+  m.def("conv_stem_wrw", [](at::Tensor dy, at::Tensor x4) {
+    TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                x4.is_contiguous(at::MemoryFormat::ChannelsLast));
+    TORCH_CHECK(dy.scalar_type() == at::kBFloat16 &&
+                x4.scalar_type() == at::kBFloat16 && x4.size(1) == 4);
+    int N = x4.size(0), Hp = x4.size(2), Wp = x4.size(3);
+    int Cout = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+    long M = (long)N * OH * OW;
+    int split = tfosr_wrw2_split(Cout, 32, M);
+    long K = 7L * 32;
+    auto ws = at::empty({(long)split, (long)Cout * K},
+                        x4.options().dtype(at::kFloat));
+    auto dW = at::empty({(long)Cout, K}, x4.options().dtype(at::kFloat));
+    auto guard = at::zeros({64}, x4.options());
+    tfosr_conv_wrw2(dy.data_ptr(), x4.data_ptr(), guard.data_ptr(),
+                    ws.data_ptr<float>(), dW.data_ptr<float>(), N, Hp, Wp,
+                    /*Cin=*/32, Cout, OH, OW, /*R=*/7, /*S_f=*/1,
+                    /*stride=*/2, /*P=*/0, split, cur_stream());
     return dW;
   });
   m.def("conv_wrw", [](at::Tensor dy, at::Tensor x, long R, long S, long P) {

@@ -44,7 +44,7 @@ __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
     const bf16_t* __restrict__ X, const bf16_t* __restrict__ W9,
     const bf16_t* __restrict__ guard, OT* __restrict__ C,
     int Nn, int H, int Wd, int Cin, int Cout, int OH, int OW, int S, int P,
-    int taps, int fw) {
+    int taps, int fw, int pixst) {
   constexpr int BM = WRG * MI * 16;
   constexpr int BN = WCG * 64;
   constexpr int ABYTES = BM * C3_BK * 2;
@@ -126,7 +126,7 @@ __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
       const char* src;
       if (ok && ih < H && iw < Wd) {
         long pix = a_pix[u] + (long)ih * Wd + iw;
-        src = (const char*)X + (pix * Cin + cin0) * 2 + a_col[u];
+        src = (const char*)X + (pix * (long)pixst + cin0) * 2 + a_col[u];
       } else {
         src = (const char*)guard + a_col[u];
       }
@@ -211,7 +211,7 @@ template <int D, bool ACC>
 static void launch_conv(const void* X, const void* W9, const void* guard,
                         void* Y, int out_bf16, int N, int H, int W, int Cin,
                         int Cout, int OH, int OW, int S, int P, int taps,
-                        int fw, hipStream_t s) {
+                        int fw, int pixst, hipStream_t s) {
   const long M = (long)N * OH * OW;
   if (Cout >= 192) {
     int ntm = (int)((M + 255) / 256), ntn = (Cout + 255) / 256;
@@ -220,12 +220,12 @@ static void launch_conv(const void* X, const void* W9, const void* guard,
       hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 2, 4, 8, D, ACC>), grid,
                          dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)W9,
                          (const bf16_t*)guard, (bf16_t*)Y, N, H, W, Cin, Cout,
-                         OH, OW, S, P, taps, fw);
+                         OH, OW, S, P, taps, fw, pixst);
     else
       hipLaunchKernelGGL((conv3x3_kernel<float, 2, 4, 8, D, ACC>), grid,
                          dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)W9,
                          (const bf16_t*)guard, (float*)Y, N, H, W, Cin, Cout,
-                         OH, OW, S, P, taps, fw);
+                         OH, OW, S, P, taps, fw, pixst);
     return;
   }
   int ntm = (int)((M + 255) / 256), ntn = (Cout + 127) / 128;
@@ -234,12 +234,12 @@ static void launch_conv(const void* X, const void* W9, const void* guard,
     hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 4, 2, 4, D, ACC>), grid,
                        dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)W9,
                        (const bf16_t*)guard, (bf16_t*)Y, N, H, W, Cin, Cout,
-                       OH, OW, S, P, taps, fw);
+                       OH, OW, S, P, taps, fw, pixst);
   else
     hipLaunchKernelGGL((conv3x3_kernel<float, 4, 2, 4, D, ACC>), grid,
                        dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)W9,
                        (const bf16_t*)guard, (float*)Y, N, H, W, Cin, Cout,
-                       OH, OW, S, P, taps, fw);
+                       OH, OW, S, P, taps, fw, pixst);
 }
 
 extern "C" {
@@ -248,7 +248,7 @@ void tfosr_conv3x3(const void* X, const void* W9, const void* guard, void* Y,
                    int out_bf16, int N, int H, int W, int Cin, int Cout,
                    int OH, int OW, int S, int P, hipStream_t s) {
   launch_conv<1, false>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH, OW,
-                        S, P, 9, 3, s);
+                        S, P, 9, 3, Cin, s);
 }
 
 // General entry: taps = fh*fw; input dilation Dil in {1, 2}
@@ -259,18 +259,33 @@ void tfosr_conv_mfma(const void* X, const void* W9, const void* guard, void* Y,
   if (Dil == 2) {
     if (accum)
       launch_conv<2, true>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH,
-                           OW, S, P, taps, fw, s);
+                           OW, S, P, taps, fw, Cin, s);
     else
       launch_conv<2, false>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH,
-                            OW, S, P, taps, fw, s);
+                            OW, S, P, taps, fw, Cin, s);
   } else {
     if (accum)
       launch_conv<1, true>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH,
-                           OW, S, P, taps, fw, s);
+                           OW, S, P, taps, fw, Cin, s);
     else
       launch_conv<1, false>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH,
-                            OW, S, P, taps, fw, s);
+                            OW, S, P, taps, fw, Cin, s);
   }
+}
+
+// ResNet stem 7x7/s2 forward over a pre-padded NHWC4 image (x4 layout:
+// [N][230][230][4] bf16, Cin 3 zero-padded to 4, spatial pad 3 baked in).
+// Each filter ROW is one uniform 32-wide K-step: "channels" = 8 px x 4 ch
+// (7 real px + 1 whose weight columns are zero), pixel stride 4 elements.
+// K = 7 x 32 = 224 vs the exact 147 — 1.52x MFMA inflation, far below the
+// 10.7x a Cin-pad-to-32 implicit GEMM would cost, and no guard loads at all.
+void tfosr_conv_stem(const void* X4, const void* W224, const void* guard,
+                     void* Y, int N, int Hp, int Wp, int Cout, int OH, int OW,
+                     hipStream_t s) {
+  // geometry mapped onto the generic kernel: taps=7 rows (fw=1 so rr=rs),
+  // S=2, P=0 (pre-padded), "Cin"=32, pixstride=4
+  launch_conv<1, false>(X4, W224, guard, Y, /*out_bf16=*/1, N, Hp, Wp, 32,
+                        Cout, OH, OW, 2, 0, 7, 1, 4, s);
 }
 
 }  // extern "C"

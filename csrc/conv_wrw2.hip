@@ -51,7 +51,7 @@ __global__ __launch_bounds__(256, 1) void conv_wrw2_kernel(
     const bf16_t* __restrict__ dy, const bf16_t* __restrict__ X,
     const bf16_t* __restrict__ guard, float* __restrict__ ws,
     int Nn, int H, int W, int Cin, int Cout, int OH, int OW, int FW,
-    int stride, int P, int steps_per_wg, long M,
+    int stride, int P, int pixst, int steps_per_wg, long M,
     unsigned long magicOW, unsigned long magicOHOW, int ntn, int split) {
   constexpr int REG = TAPS + 1;
   constexpr int REGB = 32 * 64 * 2;           // 4 KB per region
@@ -123,8 +123,8 @@ __global__ __launch_bounds__(256, 1) void conv_wrw2_kernel(
     unsigned ow = rem - oh * (unsigned)OW;
     const int ohS = (int)oh * stride, owS = (int)ow * stride;
     const char* dyrow = (const char*)dy + mm * (long)Cout * 2;
-    const char* xbase = (const char*)X + (long)n * H * W * (long)Cin * 2;
-    const long rowb = (long)W * Cin * 2, colb = (long)Cin * 2;
+    const char* xbase = (const char*)X + (long)n * H * W * (long)pixst * 2;
+    const long rowb = (long)W * pixst * 2, colb = (long)pixst * 2;
     unsigned inv = 0;
     #pragma unroll
     for (int u = 0; u < CPT; ++u) {
@@ -480,7 +480,14 @@ void tfosr_conv_wrw2(const void* dy, const void* x, const void* guard,
     hipLaunchKernelGGL((conv_wrw2_kernel<9, 2>), grid, block, 0, s,
                        (const bf16_t*)dy, (const bf16_t*)x,
                        (const bf16_t*)guard, ws, N, H, W, Cin, Cout, OH, OW,
-                       S_f, stride, P, spw, M, magicOW, magicOHOW, ntn, split);
+                       S_f, stride, P, Cin, spw, M, magicOW, magicOHOW, ntn,
+                       split);
+  else if (taps == 7)  // stem wrw over the NHWC4 padded view (pixstride 4)
+    hipLaunchKernelGGL((conv_wrw2_kernel<7, 2>), grid, block, 0, s,
+                       (const bf16_t*)dy, (const bf16_t*)x,
+                       (const bf16_t*)guard, ws, N, H, W, Cin, Cout, OH, OW,
+                       S_f, stride, P, 4, spw, M, magicOW, magicOHOW, ntn,
+                       split);
   else if (taps == 1)
     hipLaunchKernelGGL((wrw_gemm_tn_kernel<2>), grid, block, 0, s,
                        (const bf16_t*)dy, (const bf16_t*)x,

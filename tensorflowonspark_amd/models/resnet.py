@@ -16,7 +16,7 @@ import torch
 import torch.nn as nn
 
 from ..ops.modules import (Conv1x1, Conv3x3, FusedBN, FusedBNAddReLU,
-                           FusedBNReLU, FusedMaxPool2d)
+                           FusedBNReLU, FusedMaxPool2d, StemConv7x7)
 
 
 def conv3x3(cin, cout, stride=1):
@@ -125,7 +125,7 @@ class ResNet(nn.Module):
             strides = [1, 2, 2]
         else:
             self.stem = nn.Sequential(
-                nn.Conv2d(3, self.inplanes, 7, stride=2, padding=3, bias=False),
+                StemConv7x7(3, self.inplanes),
                 FusedBNReLU(self.inplanes),
                 FusedMaxPool2d(3, stride=2, padding=1))
             widths = [64, 128, 256, 512]

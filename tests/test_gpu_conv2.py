@@ -240,3 +240,52 @@ def test_fused_bottleneck_matches_eager(cfg):
     for n in g_e:
         _close(g_f[n], g_e[n], 3e-2, 2e-1, "grad " + n)
     os.environ.pop("TFOS_FUSED_BLOCK", None)
+
+
+@gpu
+@requires_gpu
+def test_stem_conv7x7():
+    """ResNet stem (7x7/s2/p3, Cin=3) on the NHWC4 implicit-GEMM path vs
+    fp32 torch: forward and weight gradient (no dgrad — first layer)."""
+    from tensorflowonspark_amd.ops.modules import StemConv7x7
+    torch.manual_seed(13)
+    m = StemConv7x7(3, 64).cuda()
+    x = torch.randn(4, 3, 96, 96, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    y = m(x)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    wr = m.weight.detach().float().cpu().requires_grad_(True)
+    yr = F.conv2d(x.float().cpu(), wr, stride=2, padding=3)
+    yr.backward(dy.float().cpu())
+    _close(y, yr, 3e-2, 3e-1, "stem fwd")
+    _close(m.weight.grad, wr.grad, 3e-2, 2.0, "stem wgrad")
+
+
+@gpu
+@requires_gpu
+def test_resnet50_full_train_step_all_tfosr():
+    """One b32 train step of the full routed model — exercises stem + fused
+    blocks + maxpool + fc + loss + optimizer together."""
+    from tensorflowonspark_amd.models import resnet50
+    from tensorflowonspark_amd.ops.modules import (BucketSGD,
+                                                   softmax_cross_entropy)
+    from tensorflowonspark_amd.parallel import DDPEngine
+    torch.manual_seed(17)
+    model = resnet50().cuda().to(memory_format=torch.channels_last)
+    model.train()
+    engine = DDPEngine(model)
+    opt = BucketSGD(engine, lr=0.05, momentum=0.9)
+    x = torch.randn(32, 3, 224, 224, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    yl = torch.randint(0, 1000, (32,), device="cuda")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        loss = softmax_cross_entropy(model(x), yl)
+    loss.backward()
+    engine.finalize_backward()
+    opt.step()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss).item()
+    g = model.stem[0].weight.grad
+    assert g is not None and torch.isfinite(g.float()).all()
