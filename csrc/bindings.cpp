@@ -55,6 +55,9 @@ void tfosr_conv_wrw2(const void*, const void*, const void*, float*, float*,
                      int, hipStream_t);
 void tfosr_conv_stem(const void*, const void*, const void*, void*, int, int,
                      int, int, int, int, hipStream_t);
+void tfosr_conv_par(const void*, const void*, const void*, void*, int, int,
+                    int, int, int, int, int, int, int, unsigned long, int,
+                    int, int, long, int, hipStream_t);
 }
 
 namespace tfosr {
@@ -446,6 +449,41 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                     ws.data_ptr<float>(), dW.data_ptr<float>(), N, H, W, Cin,
                     Cout, OH, OW, R, S_f, stride, P, split, cur_stream());
     return dW;
+  });
+  // One parity class of a stride-2 backward-data / transposed conv:
+  // out[oh*2+oh0, ow*2+ow0] = conv(x, wk_class) with the class's taps
+  // (list of (r,s) pairs). wk_class: [Cout, ntaps*Cin]. Writes into `out`
+  // (pre-allocated [N, Cout, OHr, OWr] channels_last; classes disjoint).
+  m.def("conv_par", [](at::Tensor x, at::Tensor wk, at::Tensor out,
+                       std::vector<long> taps_r, std::vector<long> taps_s,
+                       long P, bool accum) {
+    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                out.is_contiguous(at::MemoryFormat::ChannelsLast));
+    TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+                wk.scalar_type() == at::kBFloat16 &&
+                out.scalar_type() == at::kBFloat16);
+    int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+    TORCH_CHECK(Cin % 32 == 0);
+    long ntaps = (long)taps_r.size();
+    TORCH_CHECK(ntaps >= 1 && ntaps <= 8 && taps_s.size() == (size_t)ntaps);
+    TORCH_CHECK(wk.size(1) == ntaps * Cin);
+    long Cout = out.size(1), OHr = out.size(2), OWr = out.size(3);
+    // parity offsets come from the tap list: ihv = (oh*2+oh0) - P + r must be
+    // even for every class tap — caller passes oh0/ow0 via taps parity
+    // (we derive them from the first tap)
+    long oh0 = ((P - taps_r[0]) % 2 + 2) % 2;
+    long ow0 = ((P - taps_s[0]) % 2 + 2) % 2;
+    long OHs = (OHr - oh0 + 1) / 2, OWs = (OWr - ow0 + 1) / 2;
+    if (OHs <= 0 || OWs <= 0) return out;
+    unsigned long pack = 0;
+    for (long i = 0; i < ntaps; ++i)
+      pack |= ((unsigned long)((taps_r[i] << 4) | taps_s[i])) << (i * 8);
+    auto guard = at::zeros({64}, x.options());
+    tfosr_conv_par(x.data_ptr(), wk.contiguous().data_ptr(), guard.data_ptr(),
+                   out.data_ptr(), N, H, W, Cin, (int)Cout, (int)OHs,
+                   (int)OWs, (int)P, (int)ntaps, pack, (int)oh0, (int)ow0,
+                   (int)OWr, OHr * OWr, accum ? 1 : 0, cur_stream());
+    return out;
   });
   // ResNet stem: 7x7/s2/p3 conv over a pre-padded NHWC4 image
   // ([N,4,230,230] channels_last bf16, Cin 3->4 zero-padded, spatial pad 3

@@ -39,12 +39,22 @@ __device__ __forceinline__ void c3_stage16(const char* src,
 }
 
 // WRG x WCG wave grid (8 waves); per-wave output (MI*16) x 64
-template <typename OT, int WRG, int WCG, int MI, int D = 1, bool ACC = false>
+//
+// PAR mode (parity-decomposed backward-data / transposed conv): the grid
+// covers ONE output parity class — output pixel (oh*2+oh0, ow*2+ow0) over
+// subsampled dims OH x OW — with the class's tap subset supplied as a packed
+// (r<<4|s) byte list in tap_pack. Every tap of a class hits stored input
+// rows exactly (the D=2 divisibility always holds), so none of the 4x
+// dilation-zero MFMA waste of the plain D=2 formulation remains. Stores
+// land at the class's strided pixels of the REAL output (OWr/total given).
+template <typename OT, int WRG, int WCG, int MI, int D = 1, bool ACC = false,
+          bool PAR = false>
 __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
     const bf16_t* __restrict__ X, const bf16_t* __restrict__ W9,
     const bf16_t* __restrict__ guard, OT* __restrict__ C,
     int Nn, int H, int Wd, int Cin, int Cout, int OH, int OW, int S, int P,
-    int taps, int fw, int pixst) {
+    int taps, int fw, int pixst, unsigned long tap_pack, int oh0, int ow0,
+    int OWr, long OHOWr) {
   constexpr int BM = WRG * MI * 16;
   constexpr int BN = WCG * 64;
   constexpr int ABYTES = BM * C3_BK * 2;
@@ -91,8 +101,8 @@ __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
     long rem = m - n * (long)OH * OW;
     int oh = (int)(rem / OW), ow = (int)(rem - (long)(rem / OW) * OW);
     a_n[u] = ok ? n : -1;
-    a_ohS[u] = oh * S - P;
-    a_owS[u] = ow * S - P;
+    a_ohS[u] = (PAR ? oh * 2 + oh0 : oh) * S - P;
+    a_owS[u] = (PAR ? ow * 2 + ow0 : ow) * S - P;
     a_pix[u] = n * (long)H * Wd;
     a_col[u] = sl & 63;
   }
@@ -110,7 +120,10 @@ __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
   auto stage_tile = [&](int slot, int kt) {
     const int k0 = kt * C3_BK;
     const int rs = k0 / Cin;
-    const int rr = rs / fw, ss = rs - rr * fw;
+    const int rr = PAR ? (int)((tap_pack >> (rs * 8 + 4)) & 15)
+                       : rs / fw;
+    const int ss = PAR ? (int)((tap_pack >> (rs * 8)) & 15)
+                       : rs - (rs / fw) * fw;
     const int cin0 = k0 - rs * Cin;
     __attribute__((address_space(3))) char* la = lds3 + slot * SLOT;
     __attribute__((address_space(3))) char* lb = la + ABYTES;
@@ -195,6 +208,12 @@ __global__ __launch_bounds__(512, 1) void conv3x3_kernel(
     for (int r = 0; r < 4; ++r) {
       long m = tile_m + wr * MI * 16 + mi * 16 + crow0 + r;
       if (m >= M) continue;
+      if (PAR) {  // scatter to this parity class's pixels of the real output
+        long n2 = m / ((long)OH * OW);
+        long rem = m - n2 * (long)OH * OW;
+        int oh = (int)(rem / OW), ow = (int)(rem - (long)(rem / OW) * OW);
+        m = n2 * OHOWr + (long)(oh * 2 + oh0) * OWr + (ow * 2 + ow0);
+      }
       #pragma unroll
       for (int nj = 0; nj < 4; ++nj) {
         long n = tile_n + wc * 64 + nj * 16 + ccol;
@@ -220,12 +239,12 @@ static void launch_conv(const void* X, const void* W9, const void* guard,
       hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 2, 4, 8, D, ACC>), grid,
                          dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)W9,
                          (const bf16_t*)guard, (bf16_t*)Y, N, H, W, Cin, Cout,
-                         OH, OW, S, P, taps, fw, pixst);
+                         OH, OW, S, P, taps, fw, pixst, 0UL, 0, 0, OW, (long)OH * OW);
     else
       hipLaunchKernelGGL((conv3x3_kernel<float, 2, 4, 8, D, ACC>), grid,
                          dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)W9,
                          (const bf16_t*)guard, (float*)Y, N, H, W, Cin, Cout,
-                         OH, OW, S, P, taps, fw, pixst);
+                         OH, OW, S, P, taps, fw, pixst, 0UL, 0, 0, OW, (long)OH * OW);
     return;
   }
   int ntm = (int)((M + 255) / 256), ntn = (Cout + 127) / 128;
@@ -234,12 +253,12 @@ static void launch_conv(const void* X, const void* W9, const void* guard,
     hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 4, 2, 4, D, ACC>), grid,
                        dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)W9,
                        (const bf16_t*)guard, (bf16_t*)Y, N, H, W, Cin, Cout,
-                       OH, OW, S, P, taps, fw, pixst);
+                       OH, OW, S, P, taps, fw, pixst, 0UL, 0, 0, OW, (long)OH * OW);
   else
     hipLaunchKernelGGL((conv3x3_kernel<float, 4, 2, 4, D, ACC>), grid,
                        dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)W9,
                        (const bf16_t*)guard, (float*)Y, N, H, W, Cin, Cout,
-                       OH, OW, S, P, taps, fw, pixst);
+                       OH, OW, S, P, taps, fw, pixst, 0UL, 0, 0, OW, (long)OH * OW);
 }
 
 extern "C" {
@@ -271,6 +290,48 @@ void tfosr_conv_mfma(const void* X, const void* W9, const void* guard, void* Y,
       launch_conv<1, false>(X, W9, guard, Y, out_bf16, N, H, W, Cin, Cout, OH,
                             OW, S, P, taps, fw, Cin, s);
   }
+}
+
+// Parity-decomposed stride-2 backward-data / transposed-conv class launch:
+// computes the (oh0, ow0) output parity class over subsampled dims OHs x OWs
+// with ntaps class taps packed in tap_pack; writes strided pixels of the
+// real output [.., OHr, OWr] (classes are disjoint; no accumulation needed).
+void tfosr_conv_par(const void* X, const void* Wk, const void* guard, void* Y,
+                    int N, int H, int W, int Cin, int Cout, int OHs, int OWs,
+                    int P, int ntaps, unsigned long tap_pack, int oh0, int ow0,
+                    int OWr, long OHOWr, int accum, hipStream_t s) {
+  const long M = (long)N * OHs * OWs;
+  if (Cout >= 192) {
+    int ntm = (int)((M + 255) / 256), ntn = (Cout + 255) / 256;
+    dim3 grid(ntm * ntn);
+    if (accum)
+      hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 2, 4, 8, 2, true, true>),
+                         grid, dim3(512), 0, s, (const bf16_t*)X,
+                         (const bf16_t*)Wk, (const bf16_t*)guard, (bf16_t*)Y,
+                         N, H, W, Cin, Cout, OHs, OWs, 1, P, ntaps, 1, Cin,
+                         tap_pack, oh0, ow0, OWr, OHOWr);
+    else
+      hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 2, 4, 8, 2, false, true>),
+                         grid, dim3(512), 0, s, (const bf16_t*)X,
+                         (const bf16_t*)Wk, (const bf16_t*)guard, (bf16_t*)Y,
+                         N, H, W, Cin, Cout, OHs, OWs, 1, P, ntaps, 1, Cin,
+                         tap_pack, oh0, ow0, OWr, OHOWr);
+    return;
+  }
+  int ntm = (int)((M + 255) / 256), ntn = (Cout + 127) / 128;
+  dim3 grid(ntm * ntn);
+  if (accum)
+    hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 4, 2, 4, 2, true, true>), grid,
+                       dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)Wk,
+                       (const bf16_t*)guard, (bf16_t*)Y, N, H, W, Cin, Cout,
+                       OHs, OWs, 1, P, ntaps, 1, Cin, tap_pack, oh0, ow0,
+                       OWr, OHOWr);
+  else
+    hipLaunchKernelGGL((conv3x3_kernel<bf16_t, 4, 2, 4, 2, false, true>), grid,
+                       dim3(512), 0, s, (const bf16_t*)X, (const bf16_t*)Wk,
+                       (const bf16_t*)guard, (bf16_t*)Y, N, H, W, Cin, Cout,
+                       OHs, OWs, 1, P, ntaps, 1, Cin, tap_pack, oh0, ow0,
+                       OWr, OHOWr);
 }
 
 // ResNet stem 7x7/s2 forward over a pre-padded NHWC4 image (x4 layout:

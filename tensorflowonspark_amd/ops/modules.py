@@ -319,8 +319,11 @@ class _Conv1x1S2Fn(torch.autograd.Function):
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
         wT = weight.view(Cout, Cin).t().to(torch.bfloat16).contiguous()
-        # dx[h,w] = (h,w even) ? dy[h/2,w/2] @ W : 0  ==  D=2 dilated conv
-        dx = ext.conv_mfma(dy, wT, Cin, 1, 1, 1, 0, 2, x.shape[2], x.shape[3])
+        # dx[h,w] = (h,w even) ? dy[h/2,w/2] @ W : 0 — only the even-even
+        # parity class has a tap; zero-fill and launch just that class
+        dx = torch.zeros(x.shape, device=x.device, dtype=torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        _conv_parity(dy, wT.view(Cin, 1, 1, Cout), dx, 1, 0)
         import os
         if os.environ.get("TFOS_WRW", "mfma2") == "mfma2" \
                 and Cin % 8 == 0 and Cout % 8 == 0:
@@ -466,10 +469,15 @@ class _Conv3x3Fn(torch.autograd.Function):
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
         # W'[cin][r][s][cout] = W[cout][2-r][2-s][cin]
-        w9p = weight.flip(2, 3).permute(1, 2, 3, 0)             .reshape(Cin, 9 * Cout).to(torch.bfloat16).contiguous()
-        # dx = conv(dy input-dilated by S, W'), pad' = (3-1) - pad = 1
-        dx = ext.conv_mfma(dy, w9p, Cin, 3, 3, 1, 1, S,
-                           x.shape[2], x.shape[3])
+        wperm = weight.flip(2, 3).permute(1, 2, 3, 0) \
+            .to(torch.bfloat16).contiguous()
+        if S == 2:
+            dx = torch.empty(x.shape, device=x.device, dtype=torch.bfloat16) \
+                .contiguous(memory_format=torch.channels_last)
+            _conv_parity(dy, wperm, dx, 3, 1)
+        else:
+            dx = ext.conv_mfma(dy, wperm.reshape(Cin, 9 * Cout), Cin,
+                               3, 3, 1, 1, 1, x.shape[2], x.shape[3])
         import os
         wrw = os.environ.get("TFOS_WRW", "mfma2")
         if wrw == "mfma2" and Cin % 8 == 0 and Cout % 8 == 0:
@@ -518,6 +526,35 @@ class Conv3x3(nn.Module):
             self.weight.shape[1], self.weight.shape[0], self.stride)
 
 
+
+
+def _conv_parity(x, wperm, out, k, pp, accum=False):
+    """Parity-decomposed conv over a 2x-dilated input: four class launches,
+    each with only its valid taps (vs the plain D=2 kernel whose taps miss
+    the stored rows 3/4 of the time — 4x wasted MFMA work).
+
+    x: stored (undilated) input [N,KC,h,w] cl bf16; wperm: [OC, k, k, KC]
+    direct-conv weight (already flipped/swapped for the backward-data or
+    transposed-conv use); out: [N, OC, OH, OW] cl bf16 (every pixel of every
+    non-empty class gets written; for k==1 zero-fill `out` first —
+    odd-parity classes have no taps)."""
+    ext = get_ext(required=True)
+    OC = wperm.shape[0]
+    KC = wperm.shape[3]
+    for ph in (0, 1):
+        for pw in (0, 1):
+            taps = [(r, s) for r in range(k) for s in range(k)
+                    if (ph - pp + r) % 2 == 0 and (pw - pp + s) % 2 == 0]
+            if not taps or ph >= out.shape[2] or pw >= out.shape[3]:
+                continue
+            wk = torch.stack([wperm[:, r, s, :] for r, s in taps], dim=1) \
+                .reshape(OC, len(taps) * KC).contiguous()
+            ext.conv_par(x, wk, out,
+                         [r for r, _ in taps], [s for _, s in taps],
+                         pp, accum)
+    return out
+
+
 # ---------------------------------------------------------------------------
 # Transposed convolution (U-Net/DeepLab decoders) as a dilated-input conv
 # ---------------------------------------------------------------------------
@@ -536,12 +573,19 @@ class _ConvT2dFn(torch.autograd.Function):
         if x.dtype != torch.bfloat16:
             x = x.to(torch.bfloat16)
         # W'[cout][r][s][cin] = W[cin][cout][KH-1-r][KW-1-s]
-        wk = weight.flip(2, 3).permute(1, 2, 3, 0)             .reshape(Cout, KH * KW * Cin).to(torch.bfloat16).contiguous()
+        wperm = weight.flip(2, 3).permute(1, 2, 3, 0) \
+            .to(torch.bfloat16).contiguous()
         H, W = x.shape[2], x.shape[3]
         OH = (H - 1) * stride - 2 * padding + KH + output_padding
         OW = (W - 1) * stride - 2 * padding + KW + output_padding
-        y = ext.conv_mfma(x, wk, Cout, KH, KW, 1, KH - 1 - padding, stride,
-                          OH, OW)
+        y = torch.empty(x.shape[0], Cout, OH, OW, device=x.device,
+                        dtype=torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        if KH % 2 == 0 and output_padding == 0:
+            _conv_parity(x, wperm, y, KH, KH - 1 - padding)
+        else:  # odd kernels / output_padding: some classes empty -> zero base
+            y.zero_()
+            _conv_parity(x, wperm, y, KH, KH - 1 - padding)
         ctx.save_for_backward(x, weight)
         ctx.params = (stride, padding)
         return y
@@ -881,10 +925,17 @@ class _BottleneckFn(torch.autograd.Function):
         dt2, dg2, db2 = ext.bn_bwd(t2, da2, a2, k2, g2, m2, r2, True, False)
         dt2 = cl(dt2)
 
-        # conv2 (3x3, stride s): dgrad (input-dilated) + wrw
-        w9p = w2.flip(2, 3).permute(1, 2, 3, 0).reshape(C1, 9 * C2) \
+        # conv2 (3x3, stride s): dgrad (parity-decomposed for s2) + wrw
+        wperm2 = w2.flip(2, 3).permute(1, 2, 3, 0) \
             .to(torch.bfloat16).contiguous()
-        da1 = ext.conv_mfma(dt2, w9p, C1, 3, 3, 1, 1, stride, H, W)
+        if stride == 2:
+            da1 = torch.empty(N, C1, H, W, device=x.device,
+                              dtype=torch.bfloat16) \
+                .contiguous(memory_format=torch.channels_last)
+            _conv_parity(dt2, wperm2, da1, 3, 1)
+        else:
+            da1 = ext.conv_mfma(dt2, wperm2.reshape(C1, 9 * C2), C1,
+                                3, 3, 1, 1, 1, H, W)
         if wrw2:
             dw9 = ext.conv_wrw2(dt2, a1, 3, 3, stride, 1)
             dw2 = dw9.view(C2, 3, 3, C1).permute(0, 3, 1, 2).contiguous()
@@ -918,8 +969,9 @@ class _BottleneckFn(torch.autograd.Function):
             if stride == 1:
                 ext.gemm_bt_acc(as2d(dtd), wdb.t().contiguous(), as2d(dx))
             else:
-                ext.conv_mfma_acc(dtd, wdb.t().contiguous(), dx,
-                                  1, 1, 1, 0, stride)
+                # even-even parity class only; odd pixels' contribution is 0
+                _conv_parity(dtd, wdb.t().contiguous().view(Cin, 1, 1, C3),
+                             dx, 1, 0, accum=True)
             dwd = dwd.to(wd.dtype)
             dgd_, dbd_ = dgd, dbd
         else:
