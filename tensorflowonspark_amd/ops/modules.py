@@ -631,7 +631,11 @@ class ConvTranspose2dMFMA(nn.Module):
 
     def forward(self, x):
         import os
-        if x.is_cuda and self._eligible and x.dtype == torch.bfloat16 \
+        # tiny decoders (e.g. 4x4 spatial at the U-Net bottom) under-fill the
+        # 256-row implicit-GEMM tiles; those stay on the library conv
+        big = x.shape[0] * x.shape[2] * x.shape[3] * 4 >= 32768
+        if x.is_cuda and self._eligible and big \
+                and x.dtype == torch.bfloat16 \
                 and os.environ.get("TFOS_CONVT", "mfma") == "mfma" \
                 and get_ext(required=True) is not None:
             x = x.contiguous(memory_format=torch.channels_last)
