@@ -13,6 +13,23 @@ import torch.nn.functional as F
 from . import get_ext
 
 
+
+
+def _use_wrw2(k, cin, cout):
+    """Weight-gradient backend routing. TFOS_WRW: auto (default) routes each
+    shape to the measured-faster implementation (MI355X b1024, see
+    profiles/README.md — the in-tree transpose-read wrw kernel wins on
+    small-channel 3x3 and is within 0.75-1.2x of the library igemm
+    elsewhere); mfma2 forces the in-tree kernel, miopen the library."""
+    import os
+    mode = os.environ.get("TFOS_WRW", "auto")
+    if mode == "mfma2":
+        return True
+    if mode in ("miopen", "mfma"):
+        return False
+    return k == 3 and cin <= 64 and cout <= 64
+
+
 # ---------------------------------------------------------------------------
 # Fused BatchNorm + ReLU (training fwd/bwd, inference fwd)
 # ---------------------------------------------------------------------------
@@ -325,8 +342,7 @@ class _Conv1x1S2Fn(torch.autograd.Function):
             .contiguous(memory_format=torch.channels_last)
         _conv_parity(dy, wT.view(Cin, 1, 1, Cout), dx, 1, 0)
         import os
-        if os.environ.get("TFOS_WRW", "mfma2") == "mfma2" \
-                and Cin % 8 == 0 and Cout % 8 == 0:
+        if Cin % 8 == 0 and Cout % 8 == 0 and _use_wrw2(1, Cin, Cout):
             dw = ext.conv_wrw2(dy, x, 1, 1, 2, 0).view(Cout, Cin, 1, 1)
         else:
             w4 = weight.to(torch.bfloat16).contiguous(
@@ -377,9 +393,9 @@ class _Conv1x1Fn(torch.autograd.Function):
         # (round-1 TN kernel), miopen (library igemm; hipBLASLt's split-K
         # pick was ~7x slower than either).
         import os
-        wrw = os.environ.get("TFOS_WRW", "mfma2")
+        wrw = os.environ.get("TFOS_WRW", "auto")
         x4d = x2d.view(N, H, W, Cin).permute(0, 3, 1, 2)
-        if wrw == "mfma2" and Cin % 8 == 0 and Cout % 8 == 0:
+        if Cin % 8 == 0 and Cout % 8 == 0 and _use_wrw2(1, Cin, Cout):
             dw = ext.conv_wrw2(dy, x4d, 1, 1, 1, 0).view(Cout, Cin, 1, 1)
         elif wrw == "mfma" and Cin % 8 == 0 and Cout % 8 == 0:
             ext2 = get_ext(required=True)
@@ -479,8 +495,8 @@ class _Conv3x3Fn(torch.autograd.Function):
             dx = ext.conv_mfma(dy, wperm.reshape(Cin, 9 * Cout), Cin,
                                3, 3, 1, 1, 1, x.shape[2], x.shape[3])
         import os
-        wrw = os.environ.get("TFOS_WRW", "mfma2")
-        if wrw == "mfma2" and Cin % 8 == 0 and Cout % 8 == 0:
+        wrw = os.environ.get("TFOS_WRW", "auto")
+        if Cin % 8 == 0 and Cout % 8 == 0 and _use_wrw2(3, Cin, Cout):
             # transpose-read MFMA wrw kernel (stride 1 and 2)
             dw9 = ext.conv_wrw2(dy, x, 3, 3, S, 1)
             dw = dw9.view(Cout, 3, 3, Cin).permute(0, 3, 1, 2).contiguous()
@@ -907,9 +923,6 @@ class _BottleneckFn(torch.autograd.Function):
                                          True, True)
         dt3 = cl(dt3)
 
-        import os
-        wrw2 = os.environ.get("TFOS_WRW", "mfma2") == "mfma2"
-
         def lib_wrw(dyt, xt, w4, s_, p_):
             w4 = w4.to(torch.bfloat16).contiguous(
                 memory_format=torch.channels_last)
@@ -922,7 +935,8 @@ class _BottleneckFn(torch.autograd.Function):
         w3b = w3.view(C3, C2).to(torch.bfloat16)
         da2 = cl(as4d(ext.gemm_bt(as2d(dt3), w3b.t().contiguous(), True),
                       OH, OW))
-        dw3 = ext.conv_wrw2(dt3, a2, 1, 1, 1, 0).view(C3, C2, 1, 1) if wrw2 \
+        dw3 = ext.conv_wrw2(dt3, a2, 1, 1, 1, 0).view(C3, C2, 1, 1) \
+            if _use_wrw2(1, C2, C3) \
             else lib_wrw(dt3, a2, w3.view(C3, C2, 1, 1), 1, 0)
 
         # bn2+relu backward
@@ -940,7 +954,7 @@ class _BottleneckFn(torch.autograd.Function):
         else:
             da1 = ext.conv_mfma(dt2, wperm2.reshape(C1, 9 * C2), C1,
                                 3, 3, 1, 1, 1, H, W)
-        if wrw2:
+        if _use_wrw2(3, C1, C2):
             dw9 = ext.conv_wrw2(dt2, a1, 3, 3, stride, 1)
             dw2 = dw9.view(C2, 3, 3, C1).permute(0, 3, 1, 2).contiguous()
         else:
@@ -951,7 +965,8 @@ class _BottleneckFn(torch.autograd.Function):
         dt1 = cl(dt1)
 
         # conv1 (1x1) wrw
-        dw1 = ext.conv_wrw2(dt1, x, 1, 1, 1, 0).view(C1, Cin, 1, 1) if wrw2 \
+        dw1 = ext.conv_wrw2(dt1, x, 1, 1, 1, 0).view(C1, Cin, 1, 1) \
+            if _use_wrw2(1, Cin, C1) \
             else lib_wrw(dt1, x, w1.view(C1, Cin, 1, 1), 1, 0)
 
         w1b = w1.view(C1, Cin).to(torch.bfloat16)
@@ -964,7 +979,7 @@ class _BottleneckFn(torch.autograd.Function):
                                        gd, md, rd, False, False)
             dtd = cl(dtd)
             dwd = ext.conv_wrw2(dtd, x, 1, 1, stride, 0) \
-                .view(C3, Cin, 1, 1) if wrw2 \
+                .view(C3, Cin, 1, 1) if _use_wrw2(1, Cin, C3) \
                 else lib_wrw(dtd, x, wd.view(C3, Cin, 1, 1), stride, 0)
             # dx = conv1_dgrad, then downsample dgrad ACCUMULATES into it
             dx2d = ext.gemm_bt(as2d(dt1), w1b.t().contiguous(), True)
