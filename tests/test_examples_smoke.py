@@ -33,3 +33,24 @@ def test_resnet_async_example():
              "--records", "48", "--batch_size", "8", "--num_gpus", "0")
     assert r.returncode == 0, r.stderr[-2000:]
     assert "async run complete" in r.stdout
+
+
+@pytest.mark.timeout(500)
+def test_mnist_eval_example(tmp_path):
+    """Evaluator-role example (reference estimator/mnist_tf.py eval_node
+    flow): a dedicated evaluator polls checkpoints and writes eval events."""
+    import glob
+    r = _run("mnist/mnist_eval.py", "--cluster_size", "3", "--epochs", "1",
+             "--batch_size", "64", "--ckpt_every", "3", "--eval_timeout", "15",
+             "--data", str(tmp_path / "d" / "mnist.csv"),
+             "--model_dir", str(tmp_path / "m"))
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "evaluator: step" in (r.stdout + r.stderr)
+    evs = glob.glob(str(tmp_path / "m" / "eval" / "events.out.tfevents.*"))
+    assert evs, "no eval event files written"
+    from tensorflowonspark_amd import tfrecord
+    from tensorflowonspark_amd.utils import events
+    recs = list(tfrecord.tfrecord_iterator(evs[0], verify=True))
+    assert len(recs) >= 2
+    _, step, sc = events.decode_scalar_event(recs[-1])
+    assert "eval_acc" in sc
