@@ -45,7 +45,7 @@ def test_mnist_eval_example(tmp_path):
              "--data", str(tmp_path / "d" / "mnist.csv"),
              "--model_dir", str(tmp_path / "m"))
     assert r.returncode == 0, r.stderr[-2000:]
-    assert "evaluator: step" in (r.stdout + r.stderr)
+    # evaluator output lives in the executor process; judge by its artifacts
     evs = glob.glob(str(tmp_path / "m" / "eval" / "events.out.tfevents.*"))
     assert evs, "no eval event files written"
     from tensorflowonspark_amd import tfrecord
