@@ -58,6 +58,7 @@ void tfosr_conv_stem(const void*, const void*, const void*, void*, int, int,
 void tfosr_conv_par(const void*, const void*, const void*, void*, int, int,
                     int, int, int, int, int, int, int, unsigned long, int,
                     int, int, long, int, hipStream_t);
+void tfosr_pack_bf16(const void*, const void*, int, void*, long, hipStream_t);
 }
 
 namespace tfosr {
@@ -454,6 +455,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   // out[oh*2+oh0, ow*2+ow0] = conv(x, wk_class) with the class's taps
   // (list of (r,s) pairs). wk_class: [Cout, ntaps*Cin]. Writes into `out`
   // (pre-allocated [N, Cout, OHr, OWr] channels_last; classes disjoint).
+  // batched weight pack: desc buffer layout must match PackDesc in
+  // elementwise.hip (built by ops/packplan.py); one launch packs every
+  // per-step weight transform into the bf16 arena
+  m.def("pack_bf16", [](at::Tensor descs, at::Tensor cum, long ndesc,
+                        at::Tensor arena, long total) {
+    TORCH_CHECK(descs.is_cuda() && cum.is_cuda() && arena.is_cuda());
+    TORCH_CHECK(arena.scalar_type() == at::kBFloat16);
+    tfosr_pack_bf16(descs.data_ptr(), cum.data_ptr(), (int)ndesc,
+                    arena.data_ptr(), total, cur_stream());
+  });
   m.def("conv_par", [](at::Tensor x, at::Tensor wk, at::Tensor out,
                        std::vector<long> taps_r, std::vector<long> taps_s,
                        long P, bool accum) {

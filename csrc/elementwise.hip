@@ -176,3 +176,61 @@ extern "C" void tfosr_adam_step(float* p, const float* g, float* m, float* v,
   hipLaunchKernelGGL(adam_step_kernel, dim3(grid), dim3(256), 0, s,
                      p, g, m, v, lr, b1, b2, eps, wd, bc1, bc2, decoupled, n);
 }
+
+// ---------------------------------------------------------------------------
+// Batched weight pack: all per-step weight transforms (cast fp32->bf16,
+// permute, flip via negative strides, zero-pad via source-extent clamps) in
+// ONE kernel launch. ResNet-50's fused blocks otherwise issue ~300 tiny
+// permute/cast kernels per step (~3-4 ms, profiles/README backlog item 3).
+//
+// Descriptor (ints/longs packed host-side, one per output tensor):
+//   dst_off (elems into the bf16 arena), n (elems), od[4] (output dims,
+//   innermost last), ss[4] (source strides in ELEMENTS, may be negative),
+//   soff (source element offset), sv[4] (source valid extent per dim; an
+//   output coordinate >= sv[d] reads 0 — zero padding).
+// ---------------------------------------------------------------------------
+
+struct PackDesc {
+  const float* src;
+  long dst_off;
+  long soff;
+  int n;
+  int od[4];
+  int ss[4];
+  int sv[4];
+};
+
+__global__ void pack_bf16_kernel(const PackDesc* __restrict__ descs,
+                                 const long* __restrict__ cum, int ndesc,
+                                 bf16_t* __restrict__ arena, long total) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    // binary search for the descriptor owning element i
+    int lo = 0, hi = ndesc - 1;
+    while (lo < hi) {
+      int mid = (lo + hi) >> 1;
+      if (i >= cum[mid + 1]) lo = mid + 1; else hi = mid;
+    }
+    const PackDesc d = descs[lo];
+    long r = i - cum[lo];
+    int c3 = (int)(r % d.od[3]); r /= d.od[3];
+    int c2 = (int)(r % d.od[2]); r /= d.od[2];
+    int c1 = (int)(r % d.od[1]); r /= d.od[1];
+    int c0 = (int)r;
+    float v = 0.f;
+    if (c0 < d.sv[0] && c1 < d.sv[1] && c2 < d.sv[2] && c3 < d.sv[3])
+      v = d.src[d.soff + (long)c0 * d.ss[0] + (long)c1 * d.ss[1] +
+                (long)c2 * d.ss[2] + (long)c3 * d.ss[3]];
+    arena[d.dst_off + (i - cum[lo])] = (bf16_t)v;
+  }
+}
+
+extern "C" void tfosr_pack_bf16(const void* descs, const void* cum, int ndesc,
+                                void* arena, long total, hipStream_t s) {
+  int threads = 256;
+  long want = (total + threads - 1) / threads;
+  int blocks = (int)(want > 16384 ? 16384 : (want < 1 ? 1 : want));
+  hipLaunchKernelGGL(pack_bf16_kernel, dim3(blocks), dim3(threads), 0, s,
+                     (const PackDesc*)descs, (const long*)cum, ndesc,
+                     (bf16_t*)arena, total);
+}

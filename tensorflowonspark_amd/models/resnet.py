@@ -85,7 +85,8 @@ class Bottleneck(nn.Module):
                     bn2.running_mean, bn2.running_var,
                     self.conv3.weight, bn3.weight, bn3.bias,
                     bn3.running_mean, bn3.running_var,
-                    *dargs, self.stride, bn1.momentum, bn1.eps)
+                    *dargs, self.stride, bn1.momentum, bn1.eps,
+                    getattr(self, "_tfos_packs", None))
         identity = x
         out = self.bnrelu1(self.conv1(x))
         out = self.bnrelu2(self.conv2(out))
@@ -160,6 +161,11 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
+        if x.is_cuda and x.dtype == torch.bfloat16 and self.training:
+            # one batched kernel re-packs every fused-path weight when any
+            # parameter changed (vs ~300 eager permute/cast launches/step)
+            from ..ops.packplan import ensure_packed
+            ensure_packed(self, x)
         x = self.stem(x)
         x = self.stages(x)
         x = self.avgpool(x)

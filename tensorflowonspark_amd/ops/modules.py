@@ -544,6 +544,23 @@ class Conv3x3(nn.Module):
 
 
 
+def _conv_parity_packed(x, packs, prefix, out, k, pp, accum=False):
+    """Parity classes with pre-packed per-class weights (PackPlan views
+    named '<prefix>_<ph><pw>' shaped [OC, nr, ns, KC])."""
+    ext = get_ext(required=True)
+    from .packplan import _class_taps
+    for (ph, pw), (rl, sl) in _class_taps(k, pp).items():
+        if not rl or not sl or ph >= out.shape[2] or pw >= out.shape[3]:
+            continue
+        wk = packs["{}_{}{}".format(prefix, ph, pw)]
+        oc = wk.shape[-4] if wk.dim() == 4 else wk.shape[0]
+        wk2 = wk.reshape(oc, -1)
+        taps_r = [r for r in rl for _ in sl]
+        taps_s = [s2 for _ in rl for s2 in sl]
+        ext.conv_par(x, wk2, out, taps_r, taps_s, pp, accum)
+    return out
+
+
 def _conv_parity(x, wperm, out, k, pp, accum=False):
     """Parity-decomposed conv over a 2x-dilated input: four class launches,
     each with only its valid taps (vs the plain D=2 kernel whose taps miss
@@ -849,7 +866,7 @@ class _BottleneckFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w1, g1, b1, rm1, rv1, w2, g2, b2, rm2, rv2,
                 w3, g3, b3, rm3, rv3, wd, gd, bd, rmd, rvd,
-                stride, momentum, eps):
+                stride, momentum, eps, packs=None):
         ext = get_ext(required=True)
         N, Cin, H, W = x.shape
         C1, C2, C3 = w1.shape[0], w2.shape[0], w3.shape[0]
@@ -860,21 +877,29 @@ class _BottleneckFn(torch.autograd.Function):
         def as4d(t2, h, w):
             return t2.view(N, h, w, -1).permute(0, 3, 1, 2)
 
-        w1b = w1.view(C1, Cin).to(torch.bfloat16).contiguous()
+        # packed bf16 weights: from the batched PackPlan when supplied
+        # (one kernel for the whole model per optimizer step), else inline
+        if packs is not None:
+            w1b = packs["w1b"].reshape(C1, Cin)
+            w9 = packs["w9"].reshape(C2, 9 * C1)
+            w3b = packs["w3b"].reshape(C3, C2)
+        else:
+            w1b = w1.view(C1, Cin).to(torch.bfloat16).contiguous()
+            w9 = w2.permute(0, 2, 3, 1).reshape(C2, 9 * C1) \
+                .to(torch.bfloat16).contiguous()
+            w3b = w3.view(C3, C2).to(torch.bfloat16).contiguous()
         t1 = as4d(ext.gemm_bt(as2d(x), w1b, True), H, W)
         a1, m1, r1, k1 = ext.bn_fwd_train(t1, None, g1, b1, rm1, rv1,
                                           momentum, eps, True)
-        w9 = w2.permute(0, 2, 3, 1).reshape(C2, 9 * C1) \
-            .to(torch.bfloat16).contiguous()
         t2 = ext.conv_mfma(a1, w9, C2, 3, 3, stride, 1, 1, -1, -1)
         a2, m2, r2, k2 = ext.bn_fwd_train(t2, None, g2, b2, rm2, rv2,
                                           momentum, eps, True)
         OH, OW = t2.shape[2], t2.shape[3]
-        w3b = w3.view(C3, C2).to(torch.bfloat16).contiguous()
         t3 = as4d(ext.gemm_bt(as2d(a2), w3b, True), OH, OW)
 
         if wd is not None:
-            wdb = wd.view(C3, Cin).to(torch.bfloat16).contiguous()
+            wdb = packs["wdb"].reshape(C3, Cin) if packs is not None \
+                else wd.view(C3, Cin).to(torch.bfloat16).contiguous()
             if stride == 1:
                 td = as4d(ext.gemm_bt(as2d(x), wdb, True), H, W)
             else:
@@ -893,6 +918,7 @@ class _BottleneckFn(torch.autograd.Function):
                               w2, g2, t2, a2, m2, r2, k2,
                               w3, g3, t3, m3, r3, k3, y, wd, gd, *opt)
         ctx.meta = (stride, wd is not None)
+        ctx.packs = packs
         return y
 
     @staticmethod
@@ -901,6 +927,7 @@ class _BottleneckFn(torch.autograd.Function):
         (x, w1, g1, t1, a1, m1, r1, k1, w2, g2, t2, a2, m2, r2, k2,
          w3, g3, t3, m3, r3, k3, y, wd, gd, *opt) = ctx.saved_tensors
         stride, has_down = ctx.meta
+        packs = ctx.packs
         N, Cin, H, W = x.shape
         C1, C2, C3 = w1.shape[0], w2.shape[0], w3.shape[0]
         OH, OW = t2.shape[2], t2.shape[3]
@@ -932,9 +959,9 @@ class _BottleneckFn(torch.autograd.Function):
             return dw
 
         # conv3 (1x1): dgrad + wrw
-        w3b = w3.view(C3, C2).to(torch.bfloat16)
-        da2 = cl(as4d(ext.gemm_bt(as2d(dt3), w3b.t().contiguous(), True),
-                      OH, OW))
+        w3bT = packs["w3bT"].reshape(C2, C3) if packs is not None \
+            else w3.view(C3, C2).to(torch.bfloat16).t().contiguous()
+        da2 = cl(as4d(ext.gemm_bt(as2d(dt3), w3bT, True), OH, OW))
         dw3 = ext.conv_wrw2(dt3, a2, 1, 1, 1, 0).view(C3, C2, 1, 1) \
             if _use_wrw2(1, C2, C3) \
             else lib_wrw(dt3, a2, w3.view(C3, C2, 1, 1), 1, 0)
@@ -944,16 +971,21 @@ class _BottleneckFn(torch.autograd.Function):
         dt2 = cl(dt2)
 
         # conv2 (3x3, stride s): dgrad (parity-decomposed for s2) + wrw
-        wperm2 = w2.flip(2, 3).permute(1, 2, 3, 0) \
-            .to(torch.bfloat16).contiguous()
         if stride == 2:
             da1 = torch.empty(N, C1, H, W, device=x.device,
                               dtype=torch.bfloat16) \
                 .contiguous(memory_format=torch.channels_last)
-            _conv_parity(dt2, wperm2, da1, 3, 1)
+            if packs is not None:
+                _conv_parity_packed(dt2, packs, "w9p", da1, 3, 1)
+            else:
+                wperm2 = w2.flip(2, 3).permute(1, 2, 3, 0) \
+                    .to(torch.bfloat16).contiguous()
+                _conv_parity(dt2, wperm2, da1, 3, 1)
         else:
-            da1 = ext.conv_mfma(dt2, wperm2.reshape(C1, 9 * C2), C1,
-                                3, 3, 1, 1, 1, H, W)
+            w9p = packs["w9p"].reshape(C1, 9 * C2) if packs is not None \
+                else w2.flip(2, 3).permute(1, 2, 3, 0).reshape(C1, 9 * C2) \
+                .to(torch.bfloat16).contiguous()
+            da1 = ext.conv_mfma(dt2, w9p, C1, 3, 3, 1, 1, 1, H, W)
         if _use_wrw2(3, C1, C2):
             dw9 = ext.conv_wrw2(dt2, a1, 3, 3, stride, 1)
             dw2 = dw9.view(C2, 3, 3, C1).permute(0, 3, 1, 2).contiguous()
@@ -969,7 +1001,8 @@ class _BottleneckFn(torch.autograd.Function):
             if _use_wrw2(1, Cin, C1) \
             else lib_wrw(dt1, x, w1.view(C1, Cin, 1, 1), 1, 0)
 
-        w1b = w1.view(C1, Cin).to(torch.bfloat16)
+        w1bT = packs["w1bT"].reshape(Cin, C1) if packs is not None \
+            else w1.view(C1, Cin).to(torch.bfloat16).t().contiguous()
         if has_down:
             td, idn, md, rd = opt
             # downsample path: bnd backward (no relu) then conv dgrad
@@ -982,28 +1015,28 @@ class _BottleneckFn(torch.autograd.Function):
                 .view(C3, Cin, 1, 1) if _use_wrw2(1, Cin, C3) \
                 else lib_wrw(dtd, x, wd.view(C3, Cin, 1, 1), stride, 0)
             # dx = conv1_dgrad, then downsample dgrad ACCUMULATES into it
-            dx2d = ext.gemm_bt(as2d(dt1), w1b.t().contiguous(), True)
+            dx2d = ext.gemm_bt(as2d(dt1), w1bT, True)
             dx = cl(as4d(dx2d, H, W))
-            wdb = wd.view(C3, Cin).to(torch.bfloat16)
+            wdbT = packs["wdbT"].reshape(Cin, C3) if packs is not None \
+                else wd.view(C3, Cin).to(torch.bfloat16).t().contiguous()
             if stride == 1:
-                ext.gemm_bt_acc(as2d(dtd), wdb.t().contiguous(), as2d(dx))
+                ext.gemm_bt_acc(as2d(dtd), wdbT, as2d(dx))
             else:
                 # even-even parity class only; odd pixels' contribution is 0
-                _conv_parity(dtd, wdb.t().contiguous().view(Cin, 1, 1, C3),
-                             dx, 1, 0, accum=True)
+                ext.conv_par(dtd, wdbT, dx, [0], [0], 0, True)
             dwd = dwd.to(wd.dtype)
             dgd_, dbd_ = dgd, dbd
         else:
             # dx = dres + conv1_dgrad — accumulate straight into dres
             dx = dres
-            ext.gemm_bt_acc(as2d(dt1), w1b.t().contiguous(), as2d(dx))
+            ext.gemm_bt_acc(as2d(dt1), w1bT, as2d(dx))
             dwd = dgd_ = dbd_ = None
 
         return (dx, dw1.to(w1.dtype), dg1, db1, None, None,
                 dw2.to(w2.dtype), dg2, db2, None, None,
                 dw3.to(w3.dtype), dg3, db3, None, None,
                 dwd, dgd_, dbd_, None, None,
-                None, None, None)
+                None, None, None, None)
 
 
 # ---------------------------------------------------------------------------
@@ -1022,7 +1055,7 @@ class _StemConvFn(torch.autograd.Function):
     stem is the first layer; inputs don't carry grad)."""
 
     @staticmethod
-    def forward(ctx, x, weight):
+    def forward(ctx, x, weight, w224=None):
         ext = get_ext(required=True)
         N, _, H, W = x.shape
         Cout = weight.shape[0]
@@ -1030,10 +1063,11 @@ class _StemConvFn(torch.autograd.Function):
                          device=x.device) \
             .contiguous(memory_format=torch.channels_last)
         x4[:, :3, 3:H + 3, 3:W + 3] = x
-        # w224[cout][r][s*4+c]: s<7, c<3 real else zero
-        w224 = torch.zeros(Cout, 7, 8, 4, dtype=torch.bfloat16,
-                           device=x.device)
-        w224[:, :, :7, :3] = weight.permute(0, 2, 3, 1).to(torch.bfloat16)
+        if w224 is None:
+            # w224[cout][r][s*4+c]: s<7, c<3 real else zero
+            w224 = torch.zeros(Cout, 7, 8, 4, dtype=torch.bfloat16,
+                               device=x.device)
+            w224[:, :, :7, :3] = weight.permute(0, 2, 3, 1).to(torch.bfloat16)
         y = ext.conv_stem(x4, w224.reshape(Cout, 224).contiguous(), Cout)
         ctx.save_for_backward(x4, weight)
         return y
@@ -1048,7 +1082,7 @@ class _StemConvFn(torch.autograd.Function):
         dw224 = ext.conv_stem_wrw(dy, x4)          # [Cout, 224] fp32
         Cout = weight.shape[0]
         dw = dw224.view(Cout, 7, 8, 4)[:, :, :7, :3].permute(0, 3, 1, 2)
-        return None, dw.contiguous().to(weight.dtype)
+        return None, dw.contiguous().to(weight.dtype), None
 
 
 class StemConv7x7(nn.Module):
@@ -1070,7 +1104,9 @@ class StemConv7x7(nn.Module):
                 and os.environ.get("TFOS_STEM", "mfma") == "mfma" \
                 and get_ext(required=True) is not None:
             x = x.contiguous(memory_format=torch.channels_last)
-            return _StemConvFn.apply(x, self.weight)
+            packs = getattr(self, "_tfos_packs", None)
+            return _StemConvFn.apply(x, self.weight,
+                                     packs["w224"] if packs else None)
         return F.conv2d(x, self.weight.to(x.dtype), stride=2, padding=3)
 
     def extra_repr(self):
