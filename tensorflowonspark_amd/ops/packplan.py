@@ -79,7 +79,14 @@ class PackPlan:
             .view(*[d for d in shape if True])
             for (p, shape, _s, _o, _v), off in zip(self.specs, offs)]
         self._params = [p for p, *_ in self.specs]
+        self._ptrs = [p.data_ptr() for p in self._params]
         self._versions = None
+
+    def stale_pointers(self):
+        """True if any parameter's storage moved since finalize() (e.g. a
+        DDPEngine flattened params after the plan was built) — the
+        descriptors' raw pointers would then be dangling."""
+        return any(p.data_ptr() != q for p, q in zip(self._params, self._ptrs))
 
     def run_if_stale(self):
         vers = [p._version for p in self._params]
@@ -194,7 +201,8 @@ def ensure_packed(model, x):
     if not (model.training and x.is_cuda and x.dtype == torch.bfloat16):
         return False
     import os
-    if os.environ.get("TFOS_PACK_PLAN", "on") == "off":
+    if os.environ.get("TFOS_PACK_PLAN", "on") == "off" \
+            or os.environ.get("TFOS_FUSED_BLOCK", "on") == "off":
         return False
     plan = getattr(model, "_tfos_packplan", None)
     if plan is None:
@@ -206,6 +214,13 @@ def ensure_packed(model, x):
             model._tfos_packplan_failed = True
             return False
         if plan is None:
+            model._tfos_packplan_failed = True
+            return False
+        model._tfos_packplan = plan
+    if plan.stale_pointers():  # e.g. params re-flattened: rebuild descriptors
+        try:
+            plan = build_resnet_plan(model, x.device)
+        except Exception:
             model._tfos_packplan_failed = True
             return False
         model._tfos_packplan = plan
