@@ -128,7 +128,7 @@ def main():
                             input_mode=TFCluster.InputMode.SPARK,
                             master_node="chief", num_gpus=args.num_gpus)
     cluster.train(rdd, num_epochs=args.epochs)
-    cluster.shutdown(grace_secs=5)
+    cluster.shutdown(grace_secs=30)
     print("done; model_dir:", args.model_dir, "export:", args.export_dir)
     sc.stop()
 
