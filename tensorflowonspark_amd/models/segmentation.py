@@ -148,21 +148,46 @@ class ASPP(nn.Module):
 
 
 class DilatedResNet50(ResNet):
-    """ResNet-50 with stage-4 stride replaced by dilation (output stride 16)."""
+    """ResNet-50 with stage-4 stride replaced by dilation (output stride 16).
+
+    The stage-4 convs may be the routed MFMA modules (Conv3x3/Conv1x1) —
+    those are converted too: the dilated 3x3 goes back to a library conv
+    (the in-tree kernels do not implement filter dilation), the downsample
+    1x1 just drops its stride, and the whole-block fused path is disabled
+    for these blocks (it assumes undilated convs)."""
 
     def __init__(self):
         super().__init__(Bottleneck, [3, 4, 6, 3], num_classes=1)
-        # convert last stage's stride-2 convs to dilation-2
-        for m in self.stages[3].modules():
-            if isinstance(m, nn.Conv2d) and m.stride == (2, 2):
-                m.stride = (1, 1)
-                if m.kernel_size == (3, 3):
-                    m.dilation = (2, 2)
-                    m.padding = (2, 2)
+        from ..ops.modules import Conv1x1, Conv3x3
+        for blk in self.stages[3]:
+            blk._block_fusable = False
+            blk.stride = 1
+            c2 = blk.conv2
+            if isinstance(c2, Conv3x3) and c2.stride == 2:
+                new = nn.Conv2d(c2.weight.shape[1], c2.weight.shape[0], 3,
+                                stride=1, padding=2, dilation=2, bias=False)
+                with torch.no_grad():
+                    new.weight.copy_(c2.weight)
+                blk.conv2 = new
+            elif isinstance(c2, nn.Conv2d) and c2.stride == (2, 2):
+                c2.stride = (1, 1)
+                c2.dilation = (2, 2)
+                c2.padding = (2, 2)
+            ds = blk.downsample
+            if ds is not None:
+                d0 = ds[0]
+                if isinstance(d0, Conv1x1) and d0.stride == 2:
+                    d0.stride = 1
+                    d0._s2_ok = False
+                elif isinstance(d0, nn.Conv2d) and d0.stride == (2, 2):
+                    d0.stride = (1, 1)
         del self.fc
         del self.avgpool
 
     def forward(self, x):
+        if x.is_cuda and x.dtype == torch.bfloat16 and self.training:
+            from ..ops.packplan import ensure_packed
+            ensure_packed(self, x)
         x = self.stem(x)
         return self.stages(x)
 

@@ -103,3 +103,19 @@ def test_bucket_adamw_matches_torch():
         opt.step()
     for p1, p2 in zip(m1.parameters(), m2.parameters()):
         assert torch.allclose(p1, p2, atol=1e-6), (p1 - p2).abs().max()
+
+
+def test_dilated_resnet_output_stride_16():
+    """DeepLabV3 backbone: stage-4 stride must become dilation even when the
+    convs are the routed MFMA modules (round-2 regression: the original
+    surgery only matched nn.Conv2d and silently left output stride 32)."""
+    import torch
+
+    from tensorflowonspark_amd.models.segmentation import DilatedResNet50
+    m = DilatedResNet50()
+    m.eval()
+    with torch.no_grad():
+        f = m(torch.randn(1, 3, 64, 64))
+    assert f.shape[-1] == 64 // 16, f.shape
+    # fused-block path must be off for the dilated blocks
+    assert all(not b._block_fusable for b in m.stages[3])
