@@ -52,9 +52,14 @@ class Bottleneck(nn.Module):
         # fused into the conv dgrad epilogues): requires every conv on the
         # MFMA kernels and a (Conv1x1, FusedBN) downsample
         from ..ops.modules import Conv1x1, Conv3x3, FusedBN
+        bns = [self.bnrelu1, self.bnrelu2, self.bn3] + \
+            ([downsample[1]] if downsample is not None else [])
+        uniform_bn = (len({b.eps for b in bns}) == 1
+                      and len({b.momentum for b in bns}) == 1)
         self._block_fusable = (
             isinstance(self.conv1, Conv1x1) and isinstance(self.conv2, Conv3x3)
             and isinstance(self.conv3, Conv1x1) and cin % 32 == 0
+            and uniform_bn
             and (downsample is None or
                  (isinstance(downsample[0], Conv1x1)
                   and type(downsample[1]) is FusedBN)))
