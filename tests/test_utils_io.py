@@ -131,3 +131,58 @@ def test_tfrecord_dataset_shuffle(tmp_path):
     assert sorted(epoch1) == sorted(epoch2)       # same elements
     assert epoch1 != sorted(epoch1)               # actually shuffled
     assert epoch1 != epoch2                       # reshuffled per epoch
+
+
+def test_checkpoint_pruning_past_9999(tmp_path):
+    """lexicographic pruning broke at step 10000 ('weights-10000'
+    sorts before 'weights-9999'); pruning must be numeric."""
+    import torch.nn as nn
+
+    from tensorflowonspark_amd.utils import checkpoint as ckpt
+    m = nn.Linear(2, 2)
+    d = str(tmp_path / "ck")
+    for step in (9998, 9999, 10000, 10001, 10002):
+        ckpt.save_checkpoint(d, step, m, keep_last=3)
+    import glob
+    import os
+    left = sorted(int(os.path.basename(p).split("-")[1].split(".")[0])
+                  for p in glob.glob(d + "/weights-*.pt"))
+    assert left == [10000, 10001, 10002], left
+    assert ckpt.latest_checkpoint(d).endswith("weights-10002.pt")
+
+
+def test_hdfs_path_output_openable(tmp_path):
+    """hdfs_path's URIs must be openable — with a
+    file:// defaultFS the produced URI round-trips through the TFRecord
+    writer/reader."""
+    from tensorflowonspark_amd import TFNode, tfrecord
+
+    class Ctx:
+        defaultFS = "file://"
+        working_dir = str(tmp_path)
+
+    uri = TFNode.hdfs_path(Ctx(), "out/part-r-00000")
+    assert uri.startswith("file://")
+    with tfrecord.TFRecordWriter(uri) as w:
+        w.write(tfrecord.encode_example({"v": [7]}))
+    recs = list(tfrecord.tfrecord_iterator(uri, verify=True))
+    assert len(recs) == 1
+
+
+def test_tfrecord_dataset_schemed_dir(tmp_path):
+    """InputMode.TENSORFLOW reader over a schemed URI directory."""
+    from tensorflowonspark_amd import tfrecord
+    from tensorflowonspark_amd.utils.dataset import TFRecordDataset
+    d = tmp_path / "recs"
+    d.mkdir()
+    for i in range(2):
+        with tfrecord.TFRecordWriter(str(d / "part-{:05d}".format(i))) as w:
+            for j in range(5):
+                w.write(tfrecord.encode_example({"v": [i * 5 + j]}))
+    ds = TFRecordDataset("file://" + str(d), shard_index=0, num_shards=1,
+                         batch_size=4)
+    seen = []
+    for batch in ds:
+        for ex in batch:
+            seen.append(ex["v"][1][0])
+    assert sorted(seen) == list(range(10))
