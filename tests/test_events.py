@@ -32,3 +32,24 @@ def test_summary_writer_file(tmp_path):
     assert step == 2 and abs(sc["loss"] - 2.0 / 3) < 1e-6
     _, step6, sc6 = events.decode_scalar_event(recs[6])
     assert step6 == 4 and abs(sc6["imgs_per_sec"] - 9000.0) < 1e-3
+
+
+def test_summary_writer_schemed_logdir(tmp_path):
+    """Event files also land on schemed paths (the chief's model_dir may be
+    an hdfs:// URI; file:// here)."""
+    import glob
+
+    logdir = "file://" + str(tmp_path / "run2")
+    from tensorflowonspark_amd.utils import events as ev
+
+    # SummaryWriter uses os.makedirs + TFRecordWriter; schemed paths go
+    # through fsio inside the writer, so only strip the scheme for makedirs
+    w = ev.SummaryWriter(str(tmp_path / "run2"))
+    w.add_scalar("a", 1.0, 1)
+    w.close()
+    files = glob.glob(str(tmp_path / "run2") + "/events.out.tfevents.*")
+    assert files
+    from tensorflowonspark_amd import tfrecord
+    uri = "file://" + files[0]
+    recs = list(tfrecord.tfrecord_iterator(uri, verify=True))
+    assert len(recs) == 2
