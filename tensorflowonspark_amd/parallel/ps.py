@@ -174,6 +174,12 @@ class PSClient:
         self.addrs = [(h, int(p)) for h, p in
                       (a.rsplit(":", 1) for a in ps_addrs)]
         self._socks = {}
+        # one conversation per SOCKET at a time; different shards proceed in
+        # parallel (AsyncSGD's overlapped round trips run on threads)
+        self._locks = [threading.Lock() for _ in self.addrs]
+
+    def shard_lock(self, shard):
+        return self._locks[shard % len(self.addrs)]
 
     def _sock(self, shard):
         if shard not in self._socks:
@@ -272,7 +278,6 @@ class AsyncSGD:
         self._threads = {}
         self._host_out = {}
         self._pin = {}
-        self._lock = threading.Lock()
         for i, bucket in enumerate(engine._buckets):
             pf = bucket.param_flat
             assert pf is not None, "AsyncSGD requires flatten_params=True"
@@ -314,7 +319,8 @@ class AsyncSGD:
         if ev is not None:
             ev.synchronize()
         grad = self._pin[i].numpy()
-        with self._lock:  # one TCP conversation per socket at a time
+        # serialize per SHARD socket; different shards overlap
+        with self.client.shard_lock(self.client._shard(i)):
             self.client.push_pull(i, grad, self._host_out[i], self.lr,
                                   self.momentum, self.weight_decay,
                                   wire=self.wire)
