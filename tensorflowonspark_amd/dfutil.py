@@ -137,9 +137,15 @@ def saveAsTFRecords(df, output_dir):
                 w.write(rec)
 
 
-def loadTFRecords(sc, input_dir, binary_features=None):
+def loadTFRecords(sc, input_dir, binary_features=None, schema_hint=None):
     """Load TFRecord files under ``input_dir`` (plain path or schemed URI)
-    as a DataFrame with inferred schema; records provenance in ``loadedDF``."""
+    as a DataFrame with inferred schema; records provenance in ``loadedDF``.
+
+    ``schema_hint`` (a Spark-SQL simpleString like
+    ``struct<label:bigint,image:array<double>>`` or a [(name, dtype)] list)
+    overrides inference for the hinted fields and preserves their order —
+    parity with the JVM ``DFUtil.loadTFRecords(schemaHint)``
+    (reference ``DFUtil.scala:35-55,67-110``)."""
     from .utils import fsio
     if fsio.get_scheme(input_dir):
         files = fsio.fs_glob(input_dir.rstrip("/") + "/part-*") or \
@@ -155,6 +161,18 @@ def loadTFRecords(sc, input_dir, binary_features=None):
 
     first = next(tfrecord.tfrecord_iterator(files[0]))
     schema = infer_schema(tfrecord.decode_example(first), binary_features)
+    if schema_hint is not None:
+        hinted = parse_schema(schema_hint) if isinstance(schema_hint, str) \
+            else list(schema_hint)
+        rest = dict(schema)
+        # hinted fields first (hinted order + dtypes), unhinted keep inference
+        schema = hinted + [(n, t) for n, t in schema
+                           if n not in dict(hinted)]
+        schema = [(n, dict(hinted).get(n, rest.get(n))) for n, _ in schema]
+        # a binary hint also switches the DECODER off utf-8 (like the JVM
+        # loader's BinaryType handling, DFUtil.scala:119-184)
+        binary_features = list(binary_features or []) + [
+            n for n, t in hinted if t in ("binary", "array<binary>")]
 
     def _read(it):
         for path in it:

@@ -66,3 +66,23 @@ def test_binary_features_hint(sc, tmp_path):
     r = df2.collect()[0]
     assert r[0] == b"\x00\xff"   # stays bytes
     assert r[1] == "text"        # decoded to str
+
+
+def test_load_with_schema_hint(tmp_path):
+    """schema_hint overrides inference and preserves hinted field order
+    (reference DFUtil.scala:35-55,67-110)."""
+    from tensorflowonspark_amd import dfutil
+    from tensorflowonspark_amd.local_context import LocalSparkContext
+    sc = LocalSparkContext(2)
+    df = sc.createDataFrame([(1, 2.5, "x"), (2, 3.5, "y")],
+                            ["num", "val", "txt"],
+                            ["bigint", "double", "string"])
+    out = str(tmp_path / "recs")
+    dfutil.saveAsTFRecords(df, out)
+    # hint: txt first, and store it as binary instead of string
+    df2 = dfutil.loadTFRecords(sc, out,
+                               schema_hint="struct<txt:binary,num:bigint>")
+    assert df2.columns[0] == "txt" and df2.columns[1] == "num"
+    rows = sorted(df2.collect(), key=lambda r: r[1])
+    assert rows[0][1] == 1 and isinstance(rows[0][0], (bytes, bytearray))
+    assert rows[0][2] == 2.5  # unhinted column keeps inferred dtype
