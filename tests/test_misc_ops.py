@@ -119,3 +119,13 @@ def test_dilated_resnet_output_stride_16():
     assert f.shape[-1] == 64 // 16, f.shape
     # fused-block path must be off for the dilated blocks
     assert all(not b._block_fusable for b in m.stages[3])
+
+
+def test_steptimer_rate():
+    from tensorflowonspark_amd.utils.metrics import StepTimer
+    t = StepTimer(batch_size=10, log_every=2)
+    assert t.rate() == 0.0
+    t.step()  # starts the window
+    t.step()
+    t.step()  # completes a window of 2 -> rate set
+    assert t.rate() > 0.0
