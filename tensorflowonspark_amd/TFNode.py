@@ -63,10 +63,16 @@ def export_saved_model(model, export_dir, is_chief=True, require_script=False):
     """
     if not is_chief:
         return None
+    import copy
+
     import torch
     os.makedirs(export_dir, exist_ok=True)
     try:
-        scripted = torch.jit.script(model)
+        # script a deepcopy: torch.jit.script destructively swaps children
+        # via __prepare_scriptable__ (the fused/MFMA modules provide plain-op
+        # clones there) — the live training model must stay untouched. The
+        # export is a serving artifact, so it is scripted in eval mode.
+        scripted = torch.jit.script(copy.deepcopy(model).eval())
         path = os.path.join(export_dir, "model.pt")
         scripted.save(path)
     except Exception as e:
@@ -80,6 +86,14 @@ def export_saved_model(model, export_dir, is_chief=True, require_script=False):
         logger.warning(
             "torch.jit.script failed (%s); exported state_dict only to %s — "
             "TFModel.transform will NOT be able to load this export", e, path)
+    finally:
+        # release the originals the __prepare_scriptable__ hooks kept alive
+        # (guards torch's id()-memoized prepare recursion against id reuse)
+        try:
+            from .ops.modules import _SCRIPT_CLONE_KEEPALIVE
+            _SCRIPT_CLONE_KEEPALIVE.clear()
+        except Exception:
+            pass
     logger.info("exported model to %s", path)
     return path
 

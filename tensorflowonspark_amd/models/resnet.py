@@ -19,6 +19,15 @@ from ..ops.modules import (Conv1x1, Conv3x3, FusedBN, FusedBNAddReLU,
                            FusedBNReLU, FusedMaxPool2d, StemConv7x7)
 
 
+@torch.jit.ignore
+def _maybe_pack_weights(model, x) -> None:
+    # one batched kernel re-packs every fused-path weight when any parameter
+    # changed (vs ~300 eager permute/cast launches per step)
+    if x.is_cuda and x.dtype == torch.bfloat16 and model.training:
+        from ..ops.packplan import ensure_packed
+        ensure_packed(model, x)
+
+
 def conv3x3(cin, cout, stride=1):
     if stride in (1, 2) and cin % 32 == 0 and cout % 32 == 0 and cout >= 64:
         # 3x3s (stride 1 and 2) route to the implicit-GEMM MFMA kernel
@@ -119,6 +128,71 @@ class BasicBlock(nn.Module):
         return self.bn2(self.conv2(out), identity)
 
 
+class _PlainBottleneck(nn.Module):
+    """Scriptable clone of Bottleneck (plain torch ops, shared weights)."""
+
+    def __init__(self, blk):
+        super().__init__()
+        from ..ops.modules import _bn_clone, _conv_clone
+        self.conv1 = _conv_clone(blk.conv1)
+        self.bn1 = _bn_clone(blk.bnrelu1)
+        self.conv2 = _conv_clone(blk.conv2)
+        self.bn2 = _bn_clone(blk.bnrelu2)
+        self.conv3 = _conv_clone(blk.conv3)
+        self.bn3 = _bn_clone(blk.bn3)
+        if blk.downsample is not None:
+            self.downsample = nn.Sequential(
+                _conv_clone(blk.downsample[0]), _bn_clone(blk.downsample[1]))
+        else:
+            self.downsample = None
+        self.train(blk.training)
+
+    def forward(self, x):
+        identity = x
+        out = torch.relu(self.bn1(self.conv1(x)))
+        out = torch.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        return torch.relu(out + identity)
+
+
+class _PlainBasicBlock(nn.Module):
+    def __init__(self, blk):
+        super().__init__()
+        from ..ops.modules import _bn_clone, _conv_clone
+        self.conv1 = _conv_clone(blk.conv1)
+        self.bn1 = _bn_clone(blk.bnrelu1)
+        self.conv2 = _conv_clone(blk.conv2)
+        self.bn2 = _bn_clone(blk.bn2)
+        if blk.downsample is not None:
+            self.downsample = nn.Sequential(
+                _conv_clone(blk.downsample[0]), _bn_clone(blk.downsample[1]))
+        else:
+            self.downsample = None
+        self.train(blk.training)
+
+    def forward(self, x):
+        identity = x
+        out = torch.relu(self.bn1(self.conv1(x)))
+        out = self.bn2(self.conv2(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        return torch.relu(out + identity)
+
+
+def _keepalive_clone(cls):
+    def prepare(self):
+        from ..ops.modules import _SCRIPT_CLONE_KEEPALIVE
+        _SCRIPT_CLONE_KEEPALIVE.append(self)
+        return cls(self)
+    return prepare
+
+
+Bottleneck.__prepare_scriptable__ = _keepalive_clone(_PlainBottleneck)
+BasicBlock.__prepare_scriptable__ = _keepalive_clone(_PlainBasicBlock)
+
+
 class ResNet(nn.Module):
     def __init__(self, block, layers, num_classes=1000, cifar_stem=False,
                  base_width=64):
@@ -166,11 +240,8 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        if x.is_cuda and x.dtype == torch.bfloat16 and self.training:
-            # one batched kernel re-packs every fused-path weight when any
-            # parameter changed (vs ~300 eager permute/cast launches/step)
-            from ..ops.packplan import ensure_packed
-            ensure_packed(self, x)
+        if not torch.jit.is_scripting():
+            _maybe_pack_weights(self, x)
         x = self.stem(x)
         x = self.stages(x)
         x = self.avgpool(x)

@@ -16,7 +16,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.modules import ConvTranspose2dMFMA, FusedBNReLU
-from .resnet import Bottleneck, ResNet
+from .resnet import Bottleneck, ResNet, _maybe_pack_weights
 
 
 class ConvBNReLU(nn.Sequential):
@@ -185,9 +185,8 @@ class DilatedResNet50(ResNet):
         del self.avgpool
 
     def forward(self, x):
-        if x.is_cuda and x.dtype == torch.bfloat16 and self.training:
-            from ..ops.packplan import ensure_packed
-            ensure_packed(self, x)
+        if not torch.jit.is_scripting():
+            _maybe_pack_weights(self, x)
         x = self.stem(x)
         return self.stages(x)
 

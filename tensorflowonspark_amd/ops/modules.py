@@ -756,6 +756,7 @@ class DenseMFMA(nn.Module):
         m.weight = self.weight
         if self.bias is not None:
             m.bias = self.bias
+        m.train(self.training)
         return m
 
 
@@ -798,6 +799,7 @@ class Conv2dIm2colMFMA(nn.Module):
         m.weight = self.weight
         if self.bias is not None:
             m.bias = self.bias
+        m.train(self.training)
         return m
 
 
@@ -1112,3 +1114,90 @@ class StemConv7x7(nn.Module):
     def extra_repr(self):
         return "3->{} 7x7 s2 (NHWC4 implicit-GEMM MFMA)".format(
             self.weight.shape[0])
+
+
+# ---------------------------------------------------------------------------
+# TorchScript export support: every fused/MFMA module (and the blocks built
+# from them) swaps to an equivalent plain-op module at scripting time, so
+# TFNode.export_saved_model produces a loadable TorchScript model for ANY
+# in-repo architecture (TFModel.transform / serving / tfosr_infer CLI).
+# ---------------------------------------------------------------------------
+
+# torch.jit.script's __prepare_scriptable__ recursion memoizes modules by
+# id(); once a replaced original is garbage-collected its id can be recycled
+# by a freshly created clone, and the memo then resolves the NEW module to
+# the OLD replacement (observed: a stem-conv clone landing where a BatchNorm
+# belonged). Keeping every original alive until the export finishes makes id
+# reuse impossible. export_saved_model drains this list.
+_SCRIPT_CLONE_KEEPALIVE = []
+
+
+def _bn_clone(src):
+    _SCRIPT_CLONE_KEEPALIVE.append(src)
+    bn = nn.BatchNorm2d(src.num_features, eps=src.eps, momentum=src.momentum)
+    with torch.no_grad():
+        bn.weight.copy_(src.weight)
+        bn.bias.copy_(src.bias)
+        bn.running_mean.copy_(src.running_mean)
+        bn.running_var.copy_(src.running_var)
+        bn.num_batches_tracked.copy_(src.num_batches_tracked)
+    bn.train(src.training)
+    return bn
+
+
+def _conv_clone(m):
+    """nn.Conv2d equivalent of Conv1x1 / Conv3x3 / StemConv7x7 (or passthrough
+    for an already-plain conv)."""
+    if isinstance(m, nn.Conv2d):
+        return m
+    _SCRIPT_CLONE_KEEPALIVE.append(m)
+    k = m.weight.shape[2]
+    stride = getattr(m, "stride", 1)
+    if isinstance(m, StemConv7x7):
+        stride, pad = 2, 3
+    else:
+        pad = 1 if k == 3 else 0
+    new = nn.Conv2d(m.weight.shape[1], m.weight.shape[0], k, stride=stride,
+                    padding=pad, bias=False)
+    new.weight = m.weight
+    new.train(m.training)
+    return new
+
+
+def _convT_clone(m):
+    if isinstance(m, nn.ConvTranspose2d):
+        return m
+    _SCRIPT_CLONE_KEEPALIVE.append(m)
+    new = nn.ConvTranspose2d(m.weight.shape[0], m.weight.shape[1],
+                             m.weight.shape[2], stride=m.stride,
+                             padding=m.padding,
+                             output_padding=m.output_padding, bias=False)
+    new.weight = m.weight
+    new.train(m.training)
+    return new
+
+
+def _fusedbn_prepare(self):
+    _SCRIPT_CLONE_KEEPALIVE.append(self)
+    bn = _bn_clone(self)
+    if self.RELU:
+        out = nn.Sequential(bn, nn.ReLU(inplace=True))
+        out.train(self.training)
+        return out
+    return bn
+
+
+# FusedBNAddReLU takes (x, res) — its parents (the residual blocks) provide
+# their own scriptable clones below, so only the one-arg variants get hooks.
+def _maxpool_prepare(self):
+    _SCRIPT_CLONE_KEEPALIVE.append(self)
+    return nn.MaxPool2d(self.k, self.s, self.p)
+
+
+FusedMaxPool2d.__prepare_scriptable__ = _maxpool_prepare
+FusedBNReLU.__prepare_scriptable__ = _fusedbn_prepare
+FusedBN.__prepare_scriptable__ = _fusedbn_prepare
+Conv1x1.__prepare_scriptable__ = lambda self: _conv_clone(self)
+Conv3x3.__prepare_scriptable__ = lambda self: _conv_clone(self)
+StemConv7x7.__prepare_scriptable__ = lambda self: _conv_clone(self)
+ConvTranspose2dMFMA.__prepare_scriptable__ = lambda self: _convT_clone(self)
