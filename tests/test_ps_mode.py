@@ -131,7 +131,7 @@ def test_async_sgd_overlap_and_bf16_wire():
         eng.finalize_backward()
         opt.step()
         assert in_flight >= 1
-    w = float(model.weight.flatten()[0])
+    w = float(model.weight.detach().flatten()[0])
     assert abs(w - 2.0) < 0.1, w
     client.stop_all()
     srv.stop()
